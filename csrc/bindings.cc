@@ -1,0 +1,145 @@
+// Python bindings for torch_cgx_amd._C.
+//
+// Surface parity with the reference pybind module
+// (/root/reference/src/ProcessGroupCGX.cc:852-857): register_layer,
+// set_quantization_bits, set_quantization_bucket_size (the reference's
+// set_quantization_bucket_size mistakenly called SetQBits — fixed here),
+// plus the ProcessGroupCGX class for the Python-side backend creator and
+// direct kernel entry points used by the GPU tests.
+#include <torch/extension.h>
+
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <torch/csrc/utils/pybind.h>
+
+#include "backend.h"
+#include "engine.h"
+
+namespace cgx {
+namespace {
+
+DType dtype_arg(const at::Tensor& t) { return dtype_of(t); }
+
+// Compress a 1-D CUDA tensor; returns the uint8 compressed buffer
+// (zero-filled alignment padding so byte comparisons are well-defined).
+at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
+                       bool stochastic, int64_t seed) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "quantize: CUDA contiguous");
+  TORCH_CHECK(bits >= 1 && bits <= 8, "quantize: bits in [1,8]");
+  TORCH_CHECK(bucket_size >= 1, "quantize: bucket_size >= 1");
+  const DType dt = dtype_arg(x);
+  const int64_t n = x.numel();
+  const int64_t bytes = buffer_size(n, dt, (int)bits, (int)bucket_size);
+  auto out = at::zeros({std::max<int64_t>(bytes, 1)},
+                       at::TensorOptions().dtype(at::kByte).device(x.device()));
+  if (n == 0) return out;
+
+  struct Blob {
+    QuantDesc d;
+    int64_t cum[2];
+  } hb;
+  hb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), n,
+                   (int32_t)bucket_size, 0};
+  hb.cum[0] = 0;
+  hb.cum[1] = (n + bucket_size - 1) / bucket_size;
+  auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
+                           at::TensorOptions().dtype(at::kByte))
+                 .to(x.device());
+  auto stream =
+      c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.device().index());
+  const char* devp = static_cast<const char*>(dev.data_ptr());
+  launch_quantize_batch(
+      reinterpret_cast<const QuantDesc*>(devp),
+      reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
+      hb.cum[1], dt, (int)bits, (uint64_t)seed, stochastic, stream.stream(),
+      bucket_size % 8 == 0);
+  return out;
+}
+
+// Decompress `comp` into `out` (1-D CUDA tensor of n elements); add=True
+// accumulates in T precision.
+void py_dequantize(at::Tensor comp, at::Tensor out, int64_t bits,
+                   int64_t bucket_size, bool add) {
+  TORCH_CHECK(comp.is_cuda() && out.is_cuda() && out.is_contiguous());
+  const DType dt = dtype_arg(out);
+  const int64_t n = out.numel();
+  if (n == 0) return;
+  struct Blob {
+    DequantDesc d;
+    int64_t cum[2];
+  } hb;
+  hb.d = DequantDesc{comp.data_ptr<uint8_t>(), out.data_ptr(), n, 0,
+                     (int32_t)bucket_size, 1, add ? 1 : 0, 0};
+  hb.cum[0] = 0;
+  hb.cum[1] = (n + 7) / 8;
+  auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
+                           at::TensorOptions().dtype(at::kByte))
+                 .to(out.device());
+  auto stream =
+      c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(out.device().index());
+  const char* devp = static_cast<const char*>(dev.data_ptr());
+  launch_dequantize_batch(
+      reinterpret_cast<const DequantDesc*>(devp),
+      reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
+      hb.cum[1], dt, (int)bits, stream.stream());
+}
+
+int64_t py_buffer_size(int64_t n, at::ScalarType st, int64_t bits,
+                       int64_t bucket_size) {
+  DType dt = st == at::kFloat ? DType::F32
+             : st == at::kHalf ? DType::F16
+                               : DType::BF16;
+  return buffer_size(n, dt, (int)bits, (int)bucket_size);
+}
+
+std::pair<std::vector<int64_t>, std::vector<int64_t>> py_partition(
+    int64_t num_elements, int64_t world_size,
+    std::vector<int64_t> layer_numels, int64_t esize) {
+  std::vector<int64_t> offs, szs;
+  Engine::partition(num_elements, (int)world_size, layer_numels, (int)esize,
+                    &offs, &szs);
+  return {offs, szs};
+}
+
+void py_register_layer(int64_t bucket_idx, int64_t layer_idx, int64_t numel,
+                       int64_t bits, int64_t bucket_size) {
+  Registry::get().register_layer((int)bucket_idx, (int)layer_idx, numel,
+                                 (int)bits, (int)bucket_size);
+}
+
+void py_set_bits(int64_t bucket_idx, int64_t layer_idx, int64_t bits) {
+  Registry::get().set_bits((int)bucket_idx, (int)layer_idx, (int)bits);
+}
+
+void py_set_bucket(int64_t bucket_idx, int64_t layer_idx,
+                   int64_t bucket_size) {
+  Registry::get().set_bucket_size((int)bucket_idx, (int)layer_idx,
+                                  (int)bucket_size);
+}
+
+}  // namespace
+}  // namespace cgx
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  py::class_<cgx::ProcessGroupCGX,
+             c10::intrusive_ptr<cgx::ProcessGroupCGX>, c10d::Backend>(
+      m, "ProcessGroupCGX")
+      .def(py::init<c10::intrusive_ptr<c10d::Store>, int, int,
+                    c10::intrusive_ptr<c10d::Backend>>(),
+           py::arg("store"), py::arg("rank"), py::arg("size"),
+           py::arg("cpu_delegate") = nullptr);
+
+  m.def("register_layer", &cgx::py_register_layer, py::arg("bucket_idx"),
+        py::arg("layer_idx"), py::arg("layer_numel"), py::arg("bits"),
+        py::arg("bucket_size"));
+  m.def("set_quantization_bits", &cgx::py_set_bits);
+  m.def("set_quantization_bucket_size", &cgx::py_set_bucket);
+  m.def("clear_registry", [] { cgx::Registry::get().clear(); });
+
+  m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
+        py::arg("bucket_size"), py::arg("stochastic") = false,
+        py::arg("seed") = 0);
+  m.def("dequantize", &cgx::py_dequantize, py::arg("comp"), py::arg("out"),
+        py::arg("bits"), py::arg("bucket_size"), py::arg("add") = false);
+  m.def("buffer_size", &cgx::py_buffer_size);
+  m.def("partition", &cgx::py_partition);
+}

@@ -1,0 +1,73 @@
+// Host-side API of the CDNA4 max-min quantization kernels.
+//
+// Wire format (see torch_cgx_amd/ops/golden.py, the single source of truth;
+// parity with reference cuda_compression_operations.cu:68-96,219-285):
+// per slice: [2*num_buckets values of T, interleaved (unit, min)]
+//            [ceil(n*bits/8) packed bytes, groups of 8 values little-endian]
+// total size align8-padded (buffer_size()).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstddef>
+#include <cstdint>
+
+namespace cgx {
+
+enum class DType : int { F32 = 0, F16 = 1, BF16 = 2 };
+
+inline int elem_size(DType d) { return d == DType::F32 ? 4 : 2; }
+
+inline int64_t align8(int64_t x) { return (x + 7) / 8 * 8; }
+
+// Compressed byte size of an n-element slice (golden.buffer_size parity).
+inline int64_t buffer_size(int64_t n, DType dt, int bits, int bucket_size) {
+  if (n == 0) return 0;
+  const int64_t nb = (n + bucket_size - 1) / bucket_size;
+  return 2 * nb * elem_size(dt) + align8((n * bits + 7) / 8);
+}
+
+// One quantize work item: compress `n` elems at `in` into `out` bytes.
+struct QuantDesc {
+  const void* in;
+  uint8_t* out;
+  int64_t n;
+  int32_t bucket;
+  int32_t pad_;
+};
+
+// One dequantize work item: decode `n` elems from `nsrc` compressed streams
+// (at in + s*src_stride for s < nsrc), summing them in T precision in stream
+// order, then write into `out` (add=1: accumulate into existing values).
+struct DequantDesc {
+  const uint8_t* in;
+  void* out;
+  int64_t n;
+  int64_t src_stride;
+  int32_t bucket;
+  int32_t nsrc;
+  int32_t add;
+  int32_t pad_;
+};
+
+// Batched launchers. descs/cum live in DEVICE memory. `cum` is the exclusive
+// prefix over per-slice work-unit counts with a trailing total:
+//   quantize: work unit = bucket,     cum[i+1]-cum[i] = ceil(n_i/bucket_i)
+//   dequant:  work unit = 8-elem group, cum[i+1]-cum[i] = ceil(n_i/8)
+// All slices in one launch share `bits` (the engine groups by bits).
+// buckets_mult8: caller guarantees every slice's bucket_size % 8 == 0 (the
+// fused single-read path); otherwise a meta pass + generic pack pass run.
+void launch_quantize_batch(const QuantDesc* descs, const int64_t* cum,
+                           int nslices, int64_t total_buckets, DType dt,
+                           int bits, uint64_t seed, bool stochastic,
+                           hipStream_t stream, bool buckets_mult8);
+
+void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
+                             int nslices, int64_t total_groups, DType dt,
+                             int bits, hipStream_t stream);
+
+// y[i] += x[i] elementwise (T precision), n elements.
+void launch_add(const void* x, void* y, int64_t n, DType dt,
+                hipStream_t stream);
+
+}  // namespace cgx

@@ -1,0 +1,495 @@
+// Implementation of the compression engine (see engine.h).
+#include "engine.h"
+
+#include <c10/util/Exception.h>
+
+#include <algorithm>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+
+namespace cgx {
+
+// ---------------------------------------------------------------------------
+// Registry
+// ---------------------------------------------------------------------------
+Registry& Registry::get() {
+  static Registry r;
+  return r;
+}
+
+void Registry::register_layer(int bucket_idx, int layer_idx, int64_t numel,
+                              int bits, int bucket_size) {
+  std::lock_guard<std::mutex> g(mu_);
+  auto it = by_idx_.find(bucket_idx);
+  if (it == by_idx_.end()) {
+    by_idx_[bucket_idx] = order_.size();
+    order_.emplace_back();
+    order_.back().idx = bucket_idx;
+    it = by_idx_.find(bucket_idx);
+  }
+  BucketInfo& b = order_[it->second];
+  if (layer_idx == 0) {  // re-registration resets the bucket
+    b.numels.clear();
+    b.cfgs.clear();
+    b.total = 0;
+  }
+  b.numels.push_back(numel);
+  LayerConfig c;
+  c.bits = bits;
+  c.bucket_size = bucket_size > 0 ? bucket_size : 512;
+  b.cfgs.push_back(c);
+  b.total += numel;
+  cursor_ = 0;
+}
+
+void Registry::set_bits(int bucket_idx, int layer_idx, int bits) {
+  std::lock_guard<std::mutex> g(mu_);
+  auto it = by_idx_.find(bucket_idx);
+  if (it == by_idx_.end()) return;
+  auto& b = order_[it->second];
+  if (layer_idx >= 0 && (size_t)layer_idx < b.cfgs.size())
+    b.cfgs[layer_idx].bits = bits;
+}
+
+void Registry::set_bucket_size(int bucket_idx, int layer_idx, int bucket_size) {
+  std::lock_guard<std::mutex> g(mu_);
+  auto it = by_idx_.find(bucket_idx);
+  if (it == by_idx_.end()) return;
+  auto& b = order_[it->second];
+  if (layer_idx >= 0 && (size_t)layer_idx < b.cfgs.size())
+    b.cfgs[layer_idx].bucket_size = bucket_size;
+}
+
+void Registry::clear() {
+  std::lock_guard<std::mutex> g(mu_);
+  order_.clear();
+  by_idx_.clear();
+  cursor_ = 0;
+}
+
+bool Registry::empty() {
+  std::lock_guard<std::mutex> g(mu_);
+  return order_.empty();
+}
+
+bool Registry::next(int64_t numel, BucketInfo* out) {
+  std::lock_guard<std::mutex> g(mu_);
+  if (order_.empty()) return false;
+  const size_t n = order_.size();
+  const size_t start = cursor_ % n;
+  // prefer the cursor position; otherwise any bucket whose total matches
+  for (size_t k = 0; k < n; k++) {
+    const size_t i = (start + k) % n;
+    if (order_[i].total == numel) {
+      *out = order_[i];
+      cursor_ = i + 1;
+      return true;
+    }
+  }
+  return false;
+}
+
+// ---------------------------------------------------------------------------
+// EngineConfig
+// ---------------------------------------------------------------------------
+static int64_t env_int(const char* name, int64_t dflt) {
+  const char* v = std::getenv(name);
+  if (!v || !*v) return dflt;
+  return std::atoll(v);
+}
+
+EngineConfig EngineConfig::from_env() {
+  EngineConfig c;
+  c.fusion_bytes = env_int("CGX_FUSION_BUFFER_SIZE_MB", 64) << 20;
+  if (c.fusion_bytes < 2048) c.fusion_bytes = 2048;
+  c.min_elems = env_int("CGX_COMPRESSION_MINIMAL_SIZE", 16);
+  c.default_bits = (int)env_int("CGX_COMPRESSION_QUANTIZATION_BITS", 32);
+  c.default_bucket = (int)env_int("CGX_COMPRESSION_BUCKET_SIZE", 512);
+  c.stochastic = env_int("CGX_STOCHASTIC_ROUNDING", 1) != 0;
+  return c;
+}
+
+// ---------------------------------------------------------------------------
+// DescRing
+// ---------------------------------------------------------------------------
+DescRing::~DescRing() {
+  for (auto& s : slots_) {
+    if (s.host) (void)hipHostFree(s.host);
+    if (s.dev) (void)hipFree(s.dev);
+    if (s.ev) (void)hipEventDestroy(s.ev);
+  }
+}
+
+void* DescRing::acquire(size_t bytes) {
+  Slot& s = slots_[cur_];
+  if (!s.ev) CGX_HIP_CHECK(hipEventCreateWithFlags(&s.ev, hipEventDisableTiming));
+  if (s.ev_recorded) CGX_HIP_CHECK(hipEventSynchronize(s.ev));
+  if (s.cap < bytes) {
+    if (s.host) CGX_HIP_CHECK(hipHostFree(s.host));
+    if (s.dev) CGX_HIP_CHECK(hipFree(s.dev));
+    size_t cap = 4096;
+    while (cap < bytes) cap *= 2;
+    CGX_HIP_CHECK(hipHostMalloc(&s.host, cap));
+    CGX_HIP_CHECK(hipMalloc(&s.dev, cap));
+    s.cap = cap;
+  }
+  return s.host;
+}
+
+void* DescRing::commit(size_t bytes, hipStream_t stream) {
+  Slot& s = slots_[cur_];
+  CGX_HIP_CHECK(
+      hipMemcpyAsync(s.dev, s.host, bytes, hipMemcpyHostToDevice, stream));
+  CGX_HIP_CHECK(hipEventRecord(s.ev, stream));
+  s.ev_recorded = true;
+  void* dev = s.dev;
+  cur_ = (cur_ + 1) % kSlots;
+  return dev;
+}
+
+// ---------------------------------------------------------------------------
+// dtype maps
+// ---------------------------------------------------------------------------
+DType dtype_of(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return DType::F32;
+    case at::kHalf: return DType::F16;
+    case at::kBFloat16: return DType::BF16;
+    default:
+      TORCH_CHECK(false, "cgx: unsupported compression dtype ", t.scalar_type());
+  }
+}
+
+ncclDataType_t nccl_dtype(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return ncclFloat32;
+    case at::kDouble: return ncclFloat64;
+    case at::kHalf: return ncclFloat16;
+    case at::kBFloat16: return ncclBfloat16;
+    case at::kInt: return ncclInt32;
+    case at::kLong: return ncclInt64;
+    case at::kChar: return ncclInt8;
+    case at::kByte: return ncclUint8;
+    case at::kBool: return ncclUint8;
+    default:
+      TORCH_CHECK(false, "cgx: unsupported dtype for RCCL ", t.scalar_type());
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Engine
+// ---------------------------------------------------------------------------
+Engine::Engine(int rank, int size) : rank_(rank), size_(size) {
+  seed_ = 0x9E3779B97F4A7C15ull ^ (0xD1B54A32D192ED03ull * (uint64_t)(rank + 1));
+}
+
+void Engine::partition(int64_t num_elements, int ws,
+                       const std::vector<int64_t>& lnumels, int esize,
+                       std::vector<int64_t>* offsets,
+                       std::vector<int64_t>* sizes) {
+  // Mirror of torch_cgx_amd.parallel.partition.partition (tested against it).
+  int64_t offset = 0;
+  size_t li = 0;
+  int64_t n_elem = lnumels.empty() ? 0 : std::min(lnumels[0], num_elements);
+  const int64_t unit = (esize == 2) ? 8 : 4;
+  int64_t remaining = num_elements;
+  for (int r = 0; r < ws; r++) {
+    const int64_t per = remaining / (ws - r);
+    int64_t cur = 0;
+    while (cur < per) {
+      if (n_elem <= per - cur) {
+        cur += n_elem;
+        li++;
+        if (li == lnumels.size()) break;
+        n_elem = std::min(lnumels[li], num_elements);
+      } else {
+        const int64_t want = per - cur;
+        const int64_t aligned =
+            std::min(((want + unit - 1) / unit) * unit, n_elem);
+        cur += aligned;
+        n_elem -= aligned;
+      }
+    }
+    remaining -= cur;
+    sizes->push_back(cur);
+    offsets->push_back(offset);
+    offset += cur;
+  }
+}
+
+uint8_t* Engine::staging(int64_t bytes) {
+  if (bytes <= 0) bytes = 1;
+  if (!staging_.defined() || staging_.numel() < bytes) {
+    int64_t cap = 1 << 20;
+    while (cap < bytes) cap *= 2;
+    staging_ = at::empty({cap}, at::TensorOptions()
+                                    .dtype(at::kByte)
+                                    .device(at::kCUDA));
+  }
+  return staging_.data_ptr<uint8_t>();
+}
+
+void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
+                          DType dt, hipStream_t stream, bool stochastic) {
+  // group by (bits, bucket%8==0)
+  std::map<std::pair<int, bool>, std::vector<const Slice*>> groups;
+  for (const auto& s : slices) {
+    if (s.n <= 0) continue;
+    groups[{s.bits, (s.bucket % 8) == 0}].push_back(&s);
+  }
+  for (auto& [key, list] : groups) {
+    const int nsl = (int)list.size();
+    const size_t desc_bytes = sizeof(QuantDesc) * nsl;
+    const size_t cum_bytes = sizeof(int64_t) * (nsl + 1);
+    char* host = (char*)ring_.acquire(desc_bytes + cum_bytes);
+    auto* qd = reinterpret_cast<QuantDesc*>(host);
+    auto* cum = reinterpret_cast<int64_t*>(host + desc_bytes);
+    cum[0] = 0;
+    for (int i = 0; i < nsl; i++) {
+      const Slice& s = *list[i];
+      qd[i] = QuantDesc{s.data, out_base + s.comp_off, s.n, s.bucket, 0};
+      cum[i + 1] = cum[i] + (s.n + s.bucket - 1) / s.bucket;
+    }
+    char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
+    launch_quantize_batch(reinterpret_cast<QuantDesc*>(dev),
+                          reinterpret_cast<int64_t*>(dev + desc_bytes), nsl,
+                          cum[nsl], dt, key.first, seed_++, stochastic, stream,
+                          key.second);
+  }
+}
+
+void Engine::run_dequant(const std::vector<Slice>& slices,
+                         const uint8_t* in_base, int64_t src_stride, int nsrc,
+                         bool add, DType dt, hipStream_t stream) {
+  std::map<int, std::vector<const Slice*>> groups;
+  for (const auto& s : slices) {
+    if (s.n <= 0) continue;
+    groups[s.bits].push_back(&s);
+  }
+  for (auto& [bits, list] : groups) {
+    const int nsl = (int)list.size();
+    const size_t desc_bytes = sizeof(DequantDesc) * nsl;
+    const size_t cum_bytes = sizeof(int64_t) * (nsl + 1);
+    char* host = (char*)ring_.acquire(desc_bytes + cum_bytes);
+    auto* dd = reinterpret_cast<DequantDesc*>(host);
+    auto* cum = reinterpret_cast<int64_t*>(host + desc_bytes);
+    cum[0] = 0;
+    for (int i = 0; i < nsl; i++) {
+      const Slice& s = *list[i];
+      dd[i] = DequantDesc{in_base + s.comp_off, s.data,    s.n, src_stride,
+                          s.bucket,             nsrc, add ? 1 : 0, 0};
+      cum[i + 1] = cum[i] + (s.n + 7) / 8;
+    }
+    char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
+    launch_dequantize_batch(reinterpret_cast<DequantDesc*>(dev),
+                            reinterpret_cast<int64_t*>(dev + desc_bytes), nsl,
+                            cum[nsl], dt, bits, stream);
+  }
+}
+
+void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
+                       ncclComm_t comm, hipStream_t stream,
+                       const EngineConfig& cfg) {
+  const int ws = size_;
+  const int es = elem_size(dt);
+  int64_t n = 0;
+  std::vector<int64_t> lnumels;
+  lnumels.reserve(views.size());
+  for (const auto& v : views) {
+    lnumels.push_back(v.numel);
+    n += v.numel;
+  }
+  if (n == 0) return;
+
+  std::vector<int64_t> offs, szs;
+  partition(n, ws, lnumels, es, &offs, &szs);
+
+  // build per-rank slice lists + compressed chunk sizes
+  std::vector<std::vector<Slice>> rs(ws);
+  std::vector<int64_t> comp(ws, 0);
+  for (int r = 0; r < ws; r++) {
+    const int64_t start = offs[r], end = offs[r] + szs[r];
+    int64_t pos = 0, coff = 0;
+    for (const auto& v : views) {
+      const int64_t lo = std::max(pos, start);
+      const int64_t hi = std::min(pos + v.numel, end);
+      if (hi > lo) {
+        rs[r].push_back(Slice{v.data + (lo - pos) * es, hi - lo, v.bits,
+                              v.bucket_size, coff});
+        coff += buffer_size(hi - lo, dt, v.bits, v.bucket_size);
+      }
+      pos += v.numel;
+      if (pos >= end) break;
+    }
+    comp[r] = coff;
+  }
+  const int64_t mycomp = comp[rank_];
+
+  int64_t send1_total = 0;
+  for (int p = 0; p < ws; p++)
+    if (p != rank_) send1_total += comp[p];
+  const int64_t recv1_total = (int64_t)(ws - 1) * mycomp;
+  uint8_t* base = staging(send1_total + recv1_total + mycomp + send1_total);
+  uint8_t* send1 = base;
+  uint8_t* recv1 = send1 + send1_total;
+  uint8_t* send2 = recv1 + recv1_total;
+  uint8_t* recv2 = send2 + mycomp;
+
+  std::vector<int64_t> peer_off(ws, 0);
+  {
+    int64_t o = 0;
+    for (int p = 0; p < ws; p++) {
+      if (p == rank_) continue;
+      peer_off[p] = o;
+      o += comp[p];
+    }
+  }
+  auto slot = [&](int p) { return p < rank_ ? p : p - 1; };
+
+  // round 1: quantize my copy of every peer's chunk
+  {
+    std::vector<Slice> all;
+    for (int p = 0; p < ws; p++) {
+      if (p == rank_) continue;
+      for (const auto& s : rs[p]) {
+        Slice t = s;
+        t.comp_off += peer_off[p];
+        all.push_back(t);
+      }
+    }
+    run_quantize(all, send1, dt, stream, cfg.stochastic);
+  }
+
+  // round 1: exchange (grouped p2p drives all xGMI links concurrently)
+  CGX_NCCL_CHECK(ncclGroupStart());
+  for (int p = 0; p < ws; p++) {
+    if (p == rank_) continue;
+    if (comp[p] > 0)
+      CGX_NCCL_CHECK(ncclSend(send1 + peer_off[p], comp[p], ncclUint8, p,
+                              comm, stream));
+    if (mycomp > 0)
+      CGX_NCCL_CHECK(ncclRecv(recv1 + (int64_t)slot(p) * mycomp, mycomp,
+                              ncclUint8, p, comm, stream));
+  }
+  CGX_NCCL_CHECK(ncclGroupEnd());
+
+  // reduce: my raw chunk += sum of (ws-1) decoded streams (peer-ascending)
+  if (mycomp > 0) {
+    run_dequant(rs[rank_], recv1, mycomp, ws - 1, /*add=*/true, dt, stream);
+    // self-quantize the reduced chunk; the same bytes go to every peer and
+    // through my own decode so all ranks end bit-identical
+    run_quantize(rs[rank_], send2, dt, stream, cfg.stochastic);
+  }
+
+  // round 2: allgather of reduced chunks
+  CGX_NCCL_CHECK(ncclGroupStart());
+  for (int p = 0; p < ws; p++) {
+    if (p == rank_) continue;
+    if (mycomp > 0)
+      CGX_NCCL_CHECK(ncclSend(send2, mycomp, ncclUint8, p, comm, stream));
+    if (comp[p] > 0)
+      CGX_NCCL_CHECK(ncclRecv(recv2 + peer_off[p], comp[p], ncclUint8, p,
+                              comm, stream));
+  }
+  CGX_NCCL_CHECK(ncclGroupEnd());
+
+  // final decode: own chunk from send2, peers' chunks from recv2
+  if (mycomp > 0)
+    run_dequant(rs[rank_], send2, 0, 1, /*add=*/false, dt, stream);
+  {
+    std::vector<Slice> all;
+    for (int p = 0; p < ws; p++) {
+      if (p == rank_) continue;
+      for (const auto& s : rs[p]) {
+        Slice t = s;
+        t.comp_off += peer_off[p];
+        all.push_back(t);
+      }
+    }
+    run_dequant(all, recv2, 0, 1, /*add=*/false, dt, stream);
+  }
+}
+
+void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
+                       hipStream_t stream) {
+  if (size_ <= 1) return;
+  TORCH_CHECK(bucket.is_contiguous(), "cgx: bucket must be contiguous");
+  const EngineConfig cfg = EngineConfig::from_env();
+  const DType dt = dtype_of(bucket);
+  const int es = elem_size(dt);
+  char* base = static_cast<char*>(bucket.data_ptr());
+  const int64_t numel = bucket.numel();
+
+  Registry::BucketInfo info;
+  const bool matched = Registry::get().next(numel, &info);
+  std::vector<LayerView> views;
+  if (matched) {
+    int64_t off = 0;
+    for (size_t i = 0; i < info.numels.size(); i++) {
+      views.push_back(LayerView{base + off * es, info.numels[i],
+                                info.cfgs[i].bits, info.cfgs[i].bucket_size});
+      off += info.numels[i];
+    }
+  } else {
+    views.push_back(LayerView{base, numel, cfg.default_bits,
+                              cfg.default_bucket});
+  }
+
+  // split into compressible layers and uncompressed merged ranges
+  std::vector<LayerView> comp_views;
+  std::vector<std::pair<char*, int64_t>> uncomp;
+  for (const auto& v : views) {
+    const bool c = v.bits <= 8 && v.numel > cfg.min_elems;
+    if (c) {
+      comp_views.push_back(v);
+    } else {
+      if (!uncomp.empty() &&
+          uncomp.back().first + uncomp.back().second * es == v.data) {
+        uncomp.back().second += v.numel;
+      } else {
+        uncomp.emplace_back(v.data, v.numel);
+      }
+    }
+  }
+
+  if (!uncomp.empty()) {
+    const ncclDataType_t ndt = nccl_dtype(bucket);
+    CGX_NCCL_CHECK(ncclGroupStart());
+    for (auto& [ptr, cnt] : uncomp) {
+      CGX_NCCL_CHECK(
+          ncclAllReduce(ptr, ptr, cnt, ndt, ncclSum, comm, stream));
+    }
+    CGX_NCCL_CHECK(ncclGroupEnd());
+  }
+
+  // tensor-fusion chunking over the compressible layers
+  const int64_t fusion_elems = std::max<int64_t>(256, cfg.fusion_bytes / es);
+  std::vector<LayerView> cur;
+  int64_t cur_n = 0;
+  auto flush = [&]() {
+    if (!cur.empty()) {
+      sra_chunk(cur, dt, comm, stream, cfg);
+      cur.clear();
+      cur_n = 0;
+    }
+  };
+  for (const auto& v : comp_views) {
+    if (v.numel >= fusion_elems) {
+      flush();
+      for (int64_t o = 0; o < v.numel; o += fusion_elems) {
+        std::vector<LayerView> w{LayerView{v.data + o * es,
+                                           std::min(fusion_elems, v.numel - o),
+                                           v.bits, v.bucket_size}};
+        sra_chunk(w, dt, comm, stream, cfg);
+      }
+    } else {
+      if (cur_n + v.numel > fusion_elems) flush();
+      cur.push_back(v);
+      cur_n += v.numel;
+    }
+  }
+  flush();
+}
+
+}  // namespace cgx

@@ -1,0 +1,167 @@
+// Compression engine: layer registry, fusion chunking, rank partitioning and
+// the compressed Scatter-Reduce-AllGather allreduce over RCCL/xGMI.
+//
+// This is the MI355X-native equivalent of the reference's
+// mpi_allreduce_operations + compressor + reducer stack
+// (/root/reference/src/mpi_allreduce_operations.cc,
+//  src/common/scatter_reduce_allgather.cc, src/common/compressor.cc):
+// one RCCL communicator bootstrapped from the c10d Store replaces the
+// MPI/SHM/NCCL communicator trio; grouped ncclSend/ncclRecv drive all 7 xGMI
+// links concurrently; batched descriptor-driven kernels replace per-layer
+// kernel launches.
+#pragma once
+
+#include <ATen/ATen.h>
+#include <rccl/rccl.h>
+
+#include <cstdint>
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
+#include "compress.h"
+
+namespace cgx {
+
+#define CGX_HIP_CHECK(cmd)                                              \
+  do {                                                                  \
+    hipError_t e_ = (cmd);                                              \
+    TORCH_CHECK(e_ == hipSuccess, "cgx HIP error: ", hipGetErrorString(e_)); \
+  } while (0)
+
+#define CGX_NCCL_CHECK(cmd)                                              \
+  do {                                                                   \
+    ncclResult_t r_ = (cmd);                                             \
+    TORCH_CHECK(r_ == ncclSuccess, "cgx RCCL error: ", ncclGetErrorString(r_)); \
+  } while (0)
+
+struct LayerConfig {
+  int bits = 32;
+  int bucket_size = 512;
+};
+
+// Process-global layer registry (parity with the reference's static
+// MPIAllReduce_Operation::RegisterLayer / Compressor config registry,
+// mpi_allreduce_operations.h:37-49, compressor.h:93-107).
+class Registry {
+ public:
+  struct BucketInfo {
+    int idx = -1;
+    std::vector<int64_t> numels;
+    std::vector<LayerConfig> cfgs;
+    int64_t total = 0;
+  };
+
+  static Registry& get();
+
+  void register_layer(int bucket_idx, int layer_idx, int64_t numel, int bits,
+                      int bucket_size);
+  void set_bits(int bucket_idx, int layer_idx, int bits);
+  void set_bucket_size(int bucket_idx, int layer_idx, int bucket_size);
+  void clear();
+  bool empty();
+
+  // Match the next allreduce call (of `numel` total elements) to a
+  // registered bucket, cycling through registration order (the reference's
+  // modulo-cursor, extractLayers, mpi_allreduce_operations.cc:257-285 —
+  // hardened: a bucket only matches if its total numel agrees, else we fall
+  // back to whole-tensor default config instead of corrupting the mapping).
+  // Returns a copy (thread safety) or nullopt-like empty BucketInfo.
+  bool next(int64_t numel, BucketInfo* out);
+
+ private:
+  std::mutex mu_;
+  std::vector<BucketInfo> order_;
+  std::unordered_map<int, size_t> by_idx_;
+  size_t cursor_ = 0;
+};
+
+// A contiguous compressible piece of the flat bucket tensor.
+struct LayerView {
+  char* data;
+  int64_t numel;
+  int bits;
+  int bucket_size;
+};
+
+struct EngineConfig {
+  int64_t fusion_bytes = 64ll << 20;
+  int64_t min_elems = 16;
+  int default_bits = 32;
+  int default_bucket = 512;
+  bool stochastic = true;
+  static EngineConfig from_env();  // re-read every bucket like the reference
+};
+
+// Pinned-host + device descriptor upload arena with an event-guarded ring so
+// batches can be queued back-to-back without host/device races.
+class DescRing {
+ public:
+  ~DescRing();
+  // Acquire a slot of >= bytes; returns host pointer to fill.
+  void* acquire(size_t bytes);
+  // Async-copy the filled slot to device, record the guard event on stream,
+  // advance the ring; returns the DEVICE pointer for kernel args.
+  void* commit(size_t bytes, hipStream_t stream);
+
+ private:
+  static constexpr int kSlots = 8;
+  struct Slot {
+    void* host = nullptr;
+    void* dev = nullptr;
+    size_t cap = 0;
+    hipEvent_t ev = nullptr;
+    bool ev_recorded = false;
+  };
+  Slot slots_[kSlots];
+  int cur_ = 0;
+};
+
+class Engine {
+ public:
+  Engine(int rank, int size);
+
+  // SUM-allreduce of a flat fp32/fp16/bf16 CUDA tensor on `stream`:
+  // registered (or default-config) layers are quantized and reduced via SRA
+  // over RCCL; non-compressible layers go through fused ncclAllReduce.
+  void allreduce(at::Tensor bucket, ncclComm_t comm, hipStream_t stream);
+
+  // Mirror of the partition walk (reference Quantizer::GetSizesAndOffsets,
+  // compressor.cc:265-299); exposed for tests via bindings.
+  static void partition(int64_t num_elements, int world_size,
+                        const std::vector<int64_t>& layer_numels, int esize,
+                        std::vector<int64_t>* offsets,
+                        std::vector<int64_t>* sizes);
+
+ private:
+  struct Slice {  // one layer piece inside one rank chunk
+    char* data;
+    int64_t n;
+    int bits;
+    int bucket;
+    int64_t comp_off;  // byte offset of this slice in the chunk's comp stream
+  };
+
+  void sra_chunk(const std::vector<LayerView>& views, DType dt,
+                 ncclComm_t comm, hipStream_t stream,
+                 const EngineConfig& cfg);
+  uint8_t* staging(int64_t bytes);
+
+  // Launch one quantize "job list", grouping slices by (bits, bucket%8==0).
+  void run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
+                    DType dt, hipStream_t stream, bool stochastic);
+  void run_dequant(const std::vector<Slice>& slices, const uint8_t* in_base,
+                   int64_t src_stride, int nsrc, bool add, DType dt,
+                   hipStream_t stream);
+
+  int rank_, size_;
+  at::Tensor staging_;
+  DescRing ring_;
+  uint64_t seed_;
+};
+
+DType dtype_of(const at::Tensor& t);
+ncclDataType_t nccl_dtype(const at::Tensor& t);
+ncclRedOp_t nccl_op(int reduce_op_enum);  // c10d::ReduceOp::RedOpType value
+
+}  // namespace cgx

@@ -1,0 +1,504 @@
+// CDNA4 (gfx950) max-min quantization kernels for the cgx backend.
+//
+// Design (MI355X-native, not a port):
+//  * wavefront = 64 lanes; one WAVE per quantization bucket — for the default
+//    bucket sizes (512/1024) each lane owns exactly 1-2 packs of 8 values,
+//    so the min/max reduction is a pure in-register __shfl_xor butterfly
+//    (no LDS) and the values stay in registers between the meta and encode
+//    phases (single HBM read).
+//  * memory-bound workload: 16-byte vectorized loads/stores wherever the
+//    slice base is 16B-aligned; grid-stride loops sized ≫ 256 CUs.
+//  * stochastic rounding via a stateless splitmix64 hash (no RNG state
+//    buffers, fully deterministic given the per-launch seed).
+//
+// Wire-format parity with the reference implementation is defined by
+// torch_cgx_amd/ops/golden.py (see reference
+// src/common/compression/cuda_compression_operations.cu:68-96,219-285).
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+
+#include "compress.h"
+
+namespace cgx {
+namespace {
+
+constexpr float kEps = 1e-10f;
+constexpr int kThreads = 256;  // 4 waves
+constexpr int kWave = 64;
+constexpr int kMaxBlocks = 4096;  // 256 CU * 8 blocks/CU * 2
+
+__device__ __forceinline__ float randf(uint64_t seed, uint64_t key) {
+  uint64_t z = seed + key * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z ^= z >> 31;
+  return static_cast<float>(z >> 40) * (1.0f / 16777216.0f);  // [0, 1)
+}
+
+template <typename To, typename From>
+__device__ __forceinline__ To bitcast(From f) {
+  static_assert(sizeof(To) == sizeof(From));
+  To t;
+  __builtin_memcpy(&t, &f, sizeof(To));
+  return t;
+}
+
+template <typename T>
+struct RawOf;
+template <>
+struct RawOf<float> {
+  using type = uint32_t;
+};
+template <>
+struct RawOf<__half> {
+  using type = uint16_t;
+};
+template <>
+struct RawOf<__hip_bfloat16> {
+  using type = uint16_t;
+};
+
+// All element math runs on raw bit patterns + fp32; conversions below are the
+// only dtype-specific pieces.  T-precision ops (decode mul/add, accumulate)
+// are expressed as fp32 ops + round-to-T, which is bitwise identical to
+// native T arithmetic because products/sums of two T values are exact in
+// fp32 for all of {fp32, fp16, bf16}.
+template <typename T>
+__device__ __forceinline__ float raw2f(uint32_t r);
+template <>
+__device__ __forceinline__ float raw2f<float>(uint32_t r) {
+  return bitcast<float>(r);
+}
+template <>
+__device__ __forceinline__ float raw2f<__half>(uint32_t r) {
+  return __half2float(bitcast<__half>(static_cast<uint16_t>(r)));
+}
+template <>
+__device__ __forceinline__ float raw2f<__hip_bfloat16>(uint32_t r) {
+  return __bfloat162float(bitcast<__hip_bfloat16>(static_cast<uint16_t>(r)));
+}
+
+template <typename T>
+__device__ __forceinline__ uint32_t f2raw(float f);
+template <>
+__device__ __forceinline__ uint32_t f2raw<float>(float f) {
+  return bitcast<uint32_t>(f);
+}
+template <>
+__device__ __forceinline__ uint32_t f2raw<__half>(float f) {
+  return bitcast<uint16_t>(__float2half(f));
+}
+template <>
+__device__ __forceinline__ uint32_t f2raw<__hip_bfloat16>(float f) {
+  return bitcast<uint16_t>(__float2bfloat16(f));
+}
+
+// 8-element vector load/store (one pack).  aligned16 is wave-uniform for the
+// quantize kernel (slice bases are pack-aligned) and per-thread for dequant.
+template <typename T>
+__device__ __forceinline__ void load8(const T* p, bool aligned16,
+                                      uint32_t (&r)[8]) {
+  if constexpr (sizeof(T) == 4) {
+    if (aligned16) {
+      int4 a = *reinterpret_cast<const int4*>(p);
+      int4 b = *reinterpret_cast<const int4*>(p + 4);
+      r[0] = a.x; r[1] = a.y; r[2] = a.z; r[3] = a.w;
+      r[4] = b.x; r[5] = b.y; r[6] = b.z; r[7] = b.w;
+      return;
+    }
+  } else {
+    if (aligned16) {
+      int4 a = *reinterpret_cast<const int4*>(p);
+      const uint32_t w[4] = {static_cast<uint32_t>(a.x),
+                             static_cast<uint32_t>(a.y),
+                             static_cast<uint32_t>(a.z),
+                             static_cast<uint32_t>(a.w)};
+#pragma unroll
+      for (int j = 0; j < 8; j++) r[j] = (w[j >> 1] >> ((j & 1) * 16)) & 0xFFFF;
+      return;
+    }
+  }
+  using R = typename RawOf<T>::type;
+  const R* q = reinterpret_cast<const R*>(p);
+#pragma unroll
+  for (int j = 0; j < 8; j++) r[j] = q[j];
+}
+
+template <typename T>
+__device__ __forceinline__ void store8(T* p, bool aligned16,
+                                       const uint32_t (&r)[8]) {
+  if constexpr (sizeof(T) == 4) {
+    if (aligned16) {
+      int4 a = {static_cast<int>(r[0]), static_cast<int>(r[1]),
+                static_cast<int>(r[2]), static_cast<int>(r[3])};
+      int4 b = {static_cast<int>(r[4]), static_cast<int>(r[5]),
+                static_cast<int>(r[6]), static_cast<int>(r[7])};
+      *reinterpret_cast<int4*>(p) = a;
+      *reinterpret_cast<int4*>(p + 4) = b;
+      return;
+    }
+  } else {
+    if (aligned16) {
+      int4 a;
+      a.x = static_cast<int>(r[0] | (r[1] << 16));
+      a.y = static_cast<int>(r[2] | (r[3] << 16));
+      a.z = static_cast<int>(r[4] | (r[5] << 16));
+      a.w = static_cast<int>(r[6] | (r[7] << 16));
+      *reinterpret_cast<int4*>(p) = a;
+      return;
+    }
+  }
+  using R = typename RawOf<T>::type;
+  R* q = reinterpret_cast<R*>(p);
+#pragma unroll
+  for (int j = 0; j < 8; j++) q[j] = static_cast<R>(r[j]);
+}
+
+// Write the low `nb` bytes of v to p (little-endian), widest aligned stores.
+__device__ __forceinline__ void store_bytes(uint8_t* p, uint64_t v, int nb) {
+  const uintptr_t a = reinterpret_cast<uintptr_t>(p);
+  if (nb == 8 && (a & 7) == 0) {
+    *reinterpret_cast<uint64_t*>(p) = v;
+    return;
+  }
+  if (nb == 4 && (a & 3) == 0) {
+    *reinterpret_cast<uint32_t*>(p) = static_cast<uint32_t>(v);
+    return;
+  }
+  if (nb == 2 && (a & 1) == 0) {
+    *reinterpret_cast<uint16_t*>(p) = static_cast<uint16_t>(v);
+    return;
+  }
+  for (int i = 0; i < nb; i++) p[i] = static_cast<uint8_t>(v >> (8 * i));
+}
+
+__device__ __forceinline__ uint64_t load_bytes(const uint8_t* p, int nb) {
+  const uintptr_t a = reinterpret_cast<uintptr_t>(p);
+  if (nb == 8 && (a & 7) == 0) return *reinterpret_cast<const uint64_t*>(p);
+  if (nb == 4 && (a & 3) == 0) return *reinterpret_cast<const uint32_t*>(p);
+  uint64_t v = 0;
+  for (int i = 0; i < nb; i++) v |= static_cast<uint64_t>(p[i]) << (8 * i);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// Quantize: one wave per bucket.  ENCODE=true requires bucket % 8 == 0 for
+// every slice (host guarantees); ENCODE=false computes meta only (the
+// arbitrary-bucket path then packs with k_pack_generic).
+// ---------------------------------------------------------------------------
+template <typename T, int BITS, bool ENCODE>
+__global__ __launch_bounds__(kThreads) void k_quantize(
+    const QuantDesc* __restrict__ descs, const int64_t* __restrict__ cum,
+    int nslices, int64_t total_buckets, uint64_t seed, int stochastic) {
+  using R = typename RawOf<T>::type;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wid =
+      static_cast<int64_t>(blockIdx.x) * (kThreads / kWave) + (threadIdx.x / kWave);
+  const int64_t nw = static_cast<int64_t>(gridDim.x) * (kThreads / kWave);
+  constexpr float divisor = static_cast<float>((1 << BITS) - 1);
+  constexpr int MAXG = 2;  // register-stashed packs per lane (bucket <= 1024)
+
+  for (int64_t b = wid; b < total_buckets; b += nw) {
+    int lo = 0, hi = nslices;
+    while (hi - lo > 1) {
+      const int mid = (lo + hi) >> 1;
+      if (cum[mid] <= b) lo = mid; else hi = mid;
+    }
+    const QuantDesc d = descs[lo];
+    const int64_t nb_slice = cum[lo + 1] - cum[lo];
+    const int64_t lb = b - cum[lo];
+    const int64_t bstart = lb * static_cast<int64_t>(d.bucket);
+    const int cur = static_cast<int>(min(static_cast<int64_t>(d.bucket), d.n - bstart));
+    const T* in = reinterpret_cast<const T*>(d.in) + bstart;
+    const bool al16 = (reinterpret_cast<uintptr_t>(in) & 15) == 0;
+    const int ngroups = (cur + 7) >> 3;
+
+    uint32_t stash[MAXG][8];
+    float lmin = INFINITY, lmax = -INFINITY;
+    for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+      uint32_t r[8];
+      const int m = min(8, cur - g * 8);
+      if (m == 8) {
+        load8<T>(in + g * 8, al16, r);
+      } else {
+        const R* q = reinterpret_cast<const R*>(in) + g * 8;
+        for (int j = 0; j < m; j++) r[j] = q[j];
+        for (int j = m; j < 8; j++) r[j] = 0;
+      }
+      if (gi < MAXG) {
+#pragma unroll
+        for (int j = 0; j < 8; j++) stash[gi][j] = r[j];
+      }
+      for (int j = 0; j < m; j++) {
+        const float f = raw2f<T>(r[j]);
+        lmin = fminf(lmin, f);
+        lmax = fmaxf(lmax, f);
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      lmin = fminf(lmin, __shfl_xor(lmin, off));
+      lmax = fmaxf(lmax, __shfl_xor(lmax, off));
+    }
+    const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
+    const float unitf = raw2f<T>(unit_raw);
+    const float minf = lmin;
+    R* meta = reinterpret_cast<R*>(d.out);
+    if (lane == 0) {
+      meta[2 * lb] = static_cast<R>(unit_raw);
+      meta[2 * lb + 1] = static_cast<R>(f2raw<T>(lmin));
+    }
+
+    if constexpr (ENCODE) {
+      uint8_t* packed =
+          reinterpret_cast<uint8_t*>(d.out) + 2 * sizeof(R) * nb_slice;
+      const int64_t num_char = (d.n * BITS + 7) >> 3;
+      const int64_t gbase = bstart >> 3;
+      const bool live = unitf >= kEps;
+      for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+        uint32_t r[8];
+        const int m = min(8, cur - g * 8);
+        if (gi < MAXG) {
+#pragma unroll
+          for (int j = 0; j < 8; j++) r[j] = stash[gi][j];
+        } else if (m == 8) {
+          load8<T>(in + g * 8, al16, r);
+        } else {
+          const R* q = reinterpret_cast<const R*>(in) + g * 8;
+          for (int j = 0; j < m; j++) r[j] = q[j];
+        }
+        uint64_t value = 0;
+        if (live) {
+          for (int j = 0; j < m; j++) {
+            const int64_t eidx = bstart + g * 8 + j;
+            const float rnd =
+                stochastic
+                    ? randf(seed, (static_cast<uint64_t>(lo) << 44) |
+                                      static_cast<uint64_t>(eidx))
+                    : 0.5f;
+            const float dd = (raw2f<T>(r[j]) - minf) / unitf + rnd;
+            const uint32_t level =
+                static_cast<uint32_t>(fminf(floorf(dd), divisor));
+            value |= static_cast<uint64_t>(level & ((1u << BITS) - 1))
+                     << (j * BITS);
+          }
+        }
+        const int64_t gb = (gbase + g) * BITS;
+        const int nbytes =
+            static_cast<int>(min(static_cast<int64_t>(BITS), num_char - gb));
+        store_bytes(packed + gb, value, nbytes);
+      }
+    }
+  }
+}
+
+// Generic pack phase for bucket % 8 != 0 (meta already computed): grid-stride
+// threads over packs, per-element bucket lookup.
+template <typename T, int BITS>
+__global__ __launch_bounds__(kThreads) void k_pack_generic(
+    const QuantDesc* __restrict__ descs, int nslices, uint64_t seed,
+    int stochastic) {
+  using R = typename RawOf<T>::type;
+  const int64_t t0 =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  constexpr float divisor = static_cast<float>((1 << BITS) - 1);
+  for (int s = 0; s < nslices; s++) {
+    const QuantDesc d = descs[s];
+    const int64_t ngroups = (d.n + 7) >> 3;
+    const int64_t nb_slice = (d.n + d.bucket - 1) / d.bucket;
+    const R* meta = reinterpret_cast<const R*>(d.out);
+    uint8_t* packed =
+        reinterpret_cast<uint8_t*>(d.out) + 2 * sizeof(R) * nb_slice;
+    const int64_t num_char = (d.n * BITS + 7) >> 3;
+    const R* in = reinterpret_cast<const R*>(d.in);
+    for (int64_t g = t0; g < ngroups; g += stride) {
+      const int m = static_cast<int>(min(static_cast<int64_t>(8), d.n - g * 8));
+      uint64_t value = 0;
+      for (int j = 0; j < m; j++) {
+        const int64_t eidx = g * 8 + j;
+        const int64_t bk = eidx / d.bucket;
+        const float unitf = raw2f<T>(meta[2 * bk]);
+        if (unitf < kEps) continue;
+        const float minf = raw2f<T>(meta[2 * bk + 1]);
+        const float rnd =
+            stochastic ? randf(seed, (static_cast<uint64_t>(s) << 44) |
+                                         static_cast<uint64_t>(eidx))
+                       : 0.5f;
+        const float dd = (raw2f<T>(in[eidx]) - minf) / unitf + rnd;
+        const uint32_t level = static_cast<uint32_t>(fminf(floorf(dd), divisor));
+        value |= static_cast<uint64_t>(level & ((1u << BITS) - 1)) << (j * BITS);
+      }
+      const int64_t gb = g * BITS;
+      const int nbytes =
+          static_cast<int>(min(static_cast<int64_t>(BITS), num_char - gb));
+      store_bytes(packed + gb, value, nbytes);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Dequantize(+multi-source accumulate): grid-stride threads over packs.
+// Sums d.nsrc compressed streams in T precision in stream order (matching the
+// CPU simulation's per-source sequential accumulate), optionally on top of
+// the existing output values (d.add).
+// ---------------------------------------------------------------------------
+template <typename T, int BITS>
+__global__ __launch_bounds__(kThreads) void k_dequant(
+    const DequantDesc* __restrict__ descs, const int64_t* __restrict__ cum,
+    int nslices, int64_t total_groups) {
+  using R = typename RawOf<T>::type;
+  const int64_t t0 =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int64_t w = t0; w < total_groups; w += stride) {
+    int lo = 0, hi = nslices;
+    while (hi - lo > 1) {
+      const int mid = (lo + hi) >> 1;
+      if (cum[mid] <= w) lo = mid; else hi = mid;
+    }
+    const DequantDesc d = descs[lo];
+    const int64_t g = w - cum[lo];
+    const int64_t nb_slice = (d.n + d.bucket - 1) / d.bucket;
+    const int64_t meta_bytes = 2 * sizeof(R) * nb_slice;
+    const int64_t num_char = (d.n * BITS + 7) >> 3;
+    const int m = static_cast<int>(min(static_cast<int64_t>(8), d.n - g * 8));
+    const int64_t gb = g * BITS;
+    const int nbytes =
+        static_cast<int>(min(static_cast<int64_t>(BITS), num_char - gb));
+    T* outp = reinterpret_cast<T*>(d.out) + g * 8;
+    const bool al16 =
+        m == 8 && (reinterpret_cast<uintptr_t>(outp) & 15) == 0;
+    const bool one_bucket = (d.bucket & 7) == 0;
+
+    uint32_t v[8];
+    bool have = d.add != 0;
+    if (have) {
+      if (al16) {
+        load8<T>(outp, true, v);
+      } else {
+        const R* q = reinterpret_cast<const R*>(outp);
+        for (int j = 0; j < m; j++) v[j] = q[j];
+      }
+    }
+    for (int s = 0; s < d.nsrc; s++) {
+      const uint8_t* src = d.in + s * d.src_stride;
+      const R* meta = reinterpret_cast<const R*>(src);
+      const uint64_t value = load_bytes(src + meta_bytes + gb, nbytes);
+      int64_t bk0 = one_bucket ? (g * 8) / d.bucket : 0;
+      for (int j = 0; j < m; j++) {
+        const int64_t bk = one_bucket ? bk0 : (g * 8 + j) / d.bucket;
+        const uint32_t lvl =
+            static_cast<uint32_t>((value >> (j * BITS)) & ((1u << BITS) - 1));
+        const float unitf = raw2f<T>(meta[2 * bk]);
+        const float minf = raw2f<T>(meta[2 * bk + 1]);
+        // decode in T precision: round(unit*lvl) then round(min + product)
+        const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
+        const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+        if (!have && s == 0) {
+          v[j] = dec;
+        } else {
+          v[j] = f2raw<T>(raw2f<T>(v[j]) + raw2f<T>(dec));
+        }
+      }
+    }
+    if (al16) {
+      store8<T>(outp, true, v);
+    } else {
+      R* q = reinterpret_cast<R*>(outp);
+      for (int j = 0; j < m; j++) q[j] = static_cast<R>(v[j]);
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kThreads) void k_add(const T* __restrict__ x,
+                                                  T* __restrict__ y,
+                                                  int64_t n) {
+  const int64_t t0 =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  using R = typename RawOf<T>::type;
+  const R* xr = reinterpret_cast<const R*>(x);
+  R* yr = reinterpret_cast<R*>(y);
+  for (int64_t i = t0; i < n; i += stride) {
+    yr[i] = static_cast<R>(f2raw<T>(raw2f<T>(xr[i]) + raw2f<T>(yr[i])));
+  }
+}
+
+inline int grid_for(int64_t work, int per_block) {
+  const int64_t blocks = (work + per_block - 1) / per_block;
+  return static_cast<int>(blocks < 1 ? 1 : (blocks > kMaxBlocks ? kMaxBlocks : blocks));
+}
+
+#define CGX_DISPATCH_BITS(BITS_VAL, ...)                        \
+  switch (BITS_VAL) {                                           \
+    case 1: { constexpr int BITS = 1; __VA_ARGS__; break; }     \
+    case 2: { constexpr int BITS = 2; __VA_ARGS__; break; }     \
+    case 3: { constexpr int BITS = 3; __VA_ARGS__; break; }     \
+    case 4: { constexpr int BITS = 4; __VA_ARGS__; break; }     \
+    case 5: { constexpr int BITS = 5; __VA_ARGS__; break; }     \
+    case 6: { constexpr int BITS = 6; __VA_ARGS__; break; }     \
+    case 7: { constexpr int BITS = 7; __VA_ARGS__; break; }     \
+    case 8: { constexpr int BITS = 8; __VA_ARGS__; break; }     \
+    default:                                                    \
+      printf("cgx: invalid bits %d\n", BITS_VAL);               \
+      abort();                                                  \
+  }
+
+#define CGX_DISPATCH_T(DT, ...)                                       \
+  switch (DT) {                                                       \
+    case DType::F32: { using T = float; __VA_ARGS__; break; }         \
+    case DType::F16: { using T = __half; __VA_ARGS__; break; }        \
+    case DType::BF16: { using T = __hip_bfloat16; __VA_ARGS__; break; } \
+  }
+
+}  // namespace
+
+void launch_quantize_batch(const QuantDesc* descs, const int64_t* cum,
+                           int nslices, int64_t total_buckets, DType dt,
+                           int bits, uint64_t seed, bool stochastic,
+                           hipStream_t stream, bool buckets_mult8) {
+  if (total_buckets <= 0 || nslices <= 0) return;
+  const int grid = grid_for(total_buckets, kThreads / kWave);
+  CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
+    if (buckets_mult8) {
+      hipLaunchKernelGGL((k_quantize<T, BITS, true>), dim3(grid),
+                         dim3(kThreads), 0, stream, descs, cum, nslices,
+                         total_buckets, seed, (int)stochastic);
+    } else {
+      hipLaunchKernelGGL((k_quantize<T, BITS, false>), dim3(grid),
+                         dim3(kThreads), 0, stream, descs, cum, nslices,
+                         total_buckets, seed, (int)stochastic);
+      hipLaunchKernelGGL((k_pack_generic<T, BITS>), dim3(grid), dim3(kThreads),
+                         0, stream, descs, nslices, seed, (int)stochastic);
+    }
+  }));
+}
+
+void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
+                             int nslices, int64_t total_groups, DType dt,
+                             int bits, hipStream_t stream) {
+  if (total_groups <= 0 || nslices <= 0) return;
+  const int grid = grid_for(total_groups, kThreads);
+  CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
+    hipLaunchKernelGGL((k_dequant<T, BITS>), dim3(grid), dim3(kThreads), 0,
+                       stream, descs, cum, nslices, total_groups);
+  }));
+}
+
+void launch_add(const void* x, void* y, int64_t n, DType dt,
+                hipStream_t stream) {
+  if (n <= 0) return;
+  const int grid = grid_for(n, kThreads);
+  CGX_DISPATCH_T(dt, {
+    hipLaunchKernelGGL((k_add<T>), dim3(grid), dim3(kThreads), 0, stream,
+                       reinterpret_cast<const T*>(x), reinterpret_cast<T*>(y),
+                       n);
+  });
+}
+
+}  // namespace cgx
