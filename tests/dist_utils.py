@@ -1,0 +1,40 @@
+"""Multi-process test harness: spawn world_size ranks over 127.0.0.1 TCP."""
+
+from __future__ import annotations
+
+import os
+import socket
+
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+
+def free_port() -> int:
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(rank, world_size, port, fn, args, backend):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch_cgx_amd  # noqa: F401  (registers "cgx")
+    dist.init_process_group(backend, rank=rank, world_size=world_size,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        fn(rank, world_size, *args)
+    finally:
+        dist.destroy_process_group()
+
+
+def run_dist(fn, world_size=2, backend="cgx", args=()):
+    """Run fn(rank, world_size, *args) on world_size processes."""
+    port = free_port()
+    # spawn, not fork: earlier tests warm torch's OpenMP pool in the pytest
+    # parent, and forking a threaded parent deadlocks DDP workers.
+    start = os.environ.get("CGX_TEST_START_METHOD", "spawn")
+    mp.start_processes(_worker,
+                       args=(world_size, port, fn, args, backend),
+                       nprocs=world_size, join=True, start_method=start)

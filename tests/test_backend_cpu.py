@@ -1,0 +1,87 @@
+"""The cgx backend on CPU tensors (gloo delegation): registration, routing,
+collectives, world_size=2 over two processes (runs without any GPU)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+
+from dist_utils import run_dist
+
+
+def _allreduce(rank, ws):
+    for size in [1, 2, 8, 128, 1024, 100000]:
+        for dtype in [torch.float16, torch.float32, torch.int32]:
+            t = torch.tensor([rank + 1] * size, dtype=dtype)
+            dist.all_reduce(t)
+            expected = torch.tensor([(ws * (ws + 1)) // 2] * size, dtype=dtype)
+            assert torch.equal(t, expected), (size, dtype)
+
+
+def test_allreduce_cpu_ws2():
+    run_dist(_allreduce, 2)
+
+
+def _collectives(rank, ws):
+    t = torch.full((16,), float(rank))
+    dist.broadcast(t, src=0)
+    assert torch.equal(t, torch.zeros(16))
+
+    t = torch.full((4,), float(rank + 1))
+    out = [torch.zeros(4) for _ in range(ws)]
+    dist.all_gather(out, t)
+    for p in range(ws):
+        assert torch.equal(out[p], torch.full((4,), float(p + 1)))
+
+    dist.barrier()
+
+    r = torch.full((8,), float(rank + 1))
+    dist.reduce(r, dst=0)
+    if rank == 0:
+        assert torch.equal(r, torch.full((8,), float(sum(range(1, ws + 1)))))
+
+    if rank == 0:
+        dist.send(torch.arange(5.0), dst=1)
+    elif rank == 1:
+        rt = torch.zeros(5)
+        dist.recv(rt, src=0)
+        assert torch.equal(rt, torch.arange(5.0))
+
+
+def test_collectives_cpu_ws2():
+    run_dist(_collectives, 2)
+
+
+def _min_max(rank, ws):
+    t = torch.tensor([float(rank + 1)] * 8)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    assert torch.equal(t, torch.full((8,), float(ws)))
+    t = torch.tensor([float(rank + 1)] * 8)
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    assert torch.equal(t, torch.full((8,), 1.0))
+
+
+def test_reduce_ops_cpu():
+    run_dist(_min_max, 2)
+
+
+def test_registry_api():
+    from torch_cgx_amd import _C
+    _C.clear_registry()
+    _C.register_layer(0, 0, 1000, 4, 512)
+    _C.register_layer(0, 1, 24, 32, 512)
+    _C.register_layer(1, 0, 5000, 4, 512)
+    _C.set_quantization_bits(0, 0, 8)
+    _C.set_quantization_bucket_size(0, 0, 1024)
+    _C.clear_registry()
+
+
+def _ws3(rank, ws):
+    t = torch.tensor([rank + 1.0] * 33)
+    dist.all_reduce(t)
+    assert torch.equal(t, torch.full((33,), 6.0))
+
+
+def test_allreduce_cpu_ws3():
+    run_dist(_ws3, 3)

@@ -1,0 +1,73 @@
+"""CPU simulation of the compressed SRA allreduce: exactness on equal inputs
+and the reference's analytic error bound (test_cgx.py:92 parity)."""
+
+import numpy as np
+import pytest
+import torch
+
+from torch_cgx_amd.parallel import sra_sim
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16])
+@pytest.mark.parametrize("bits", [2, 4, 8])
+def test_equal_inputs_exact(ws, dtype, bits):
+    for n in [8, 128, 1024, 5000]:
+        tensors = [torch.full((n,), float(3), dtype=dtype) for _ in range(ws)]
+        out = sra_sim.sra_allreduce(tensors, [n], [(bits, 512)])
+        expected = torch.full((n,), float(3 * ws), dtype=dtype)
+        for o in out:
+            assert torch.equal(o, expected), (n, bits)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+@pytest.mark.parametrize("bits", [2, 4, 8])
+@pytest.mark.parametrize("bucket", [64, 512, 2048])
+def test_error_bound_fp32(ws, bits, bucket):
+    for n in [128, 1024, 1025, 16384]:
+        arange = np.arange(-n / 2, n / 2, 1.0)
+        tensors = [torch.tensor((r + 1) * arange, dtype=torch.float32)
+                   for r in range(ws)]
+        expected = torch.tensor((ws * (ws + 1) / 2) * arange,
+                                dtype=torch.float32)
+        out = sra_sim.sra_allreduce(tensors, [n], [(bits, bucket)])
+        coef = ws * (ws + 1)
+        bound = 2 * min(bucket, n) / ((1 << bits) - 1) * coef
+        for o in out:
+            err = (o - expected).abs().max().item()
+            assert err < bound, (n, bits, bucket, err, bound)
+
+
+def test_all_ranks_bitwise_identical():
+    torch.manual_seed(0)
+    ws, n = 4, 4096
+    tensors = [torch.randn(n) for _ in range(ws)]
+    out = sra_sim.sra_allreduce(tensors, [n], [(4, 512)])
+    for o in out[1:]:
+        assert torch.equal(o, out[0])
+
+
+def test_multi_layer_mixed_config():
+    torch.manual_seed(1)
+    ws = 2
+    layers = [1000, 513, 2048]
+    cfgs = [(4, 512), (8, 64), (2, 1024)]
+    n = sum(layers)
+    tensors = [torch.randn(n) for _ in range(ws)]
+    expected = sum(tensors)
+    out = sra_sim.sra_allreduce(tensors, layers, cfgs)
+    # loose bound: sum of per-layer bounds
+    err = (out[0] - expected).abs().max().item()
+    spread = max(t.abs().max().item() for t in tensors)
+    bound = 2 * 1024 / 3 * ws * (ws + 1)  # dominated by the 2-bit layer
+    assert err < bound
+    assert torch.equal(out[0], out[1])
+
+
+def test_tiny_chunks_more_ranks_than_elems():
+    ws, n = 4, 5
+    tensors = [torch.full((n,), float(r + 1)) for r in range(ws)]
+    out = sra_sim.sra_allreduce(tensors, [n], [(4, 512)])
+    expected = torch.full((n,), float(sum(range(1, ws + 1))))
+    for o in out:
+        assert torch.allclose(o, expected)

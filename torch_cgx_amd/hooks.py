@@ -12,8 +12,6 @@ Tensors with dim <= 1 (biases, LayerNorm/BatchNorm weights) or fewer than
 quantization — the "layerwise filter".
 """
 
-from __future__ import annotations
-
 import os
 from typing import Dict, Optional
 
@@ -54,7 +52,7 @@ class CGXState:
 
 
 def _allreduce_fut(process_group: dist.ProcessGroup, tensor: torch.Tensor
-                   ) -> torch.futures.Future:
+                   ) -> torch.futures.Future[torch.Tensor]:
     group_to_use = (process_group if process_group is not None
                     else dist.group.WORLD)
     # Divide first to avoid fp16 overflow; the backend computes SUM.
@@ -64,7 +62,7 @@ def _allreduce_fut(process_group: dist.ProcessGroup, tensor: torch.Tensor
             .then(lambda fut: fut.value()[0]))
 
 
-def cgx_hook(state: CGXState, bucket) -> torch.futures.Future:
+def cgx_hook(state: CGXState, bucket) -> torch.futures.Future[torch.Tensor]:
     if state.step == 2:
         for layer_idx, tensor in enumerate(bucket.gradients()):
             bits = (state.quantization_bits if state.should_compress_(tensor)
