@@ -1,0 +1,164 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 DDP over the cgx compressed-allreduce
+backend on synthetic ImageNet-shaped data (BASELINE.json config 3).
+
+Single process:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi GPU (driver): python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Rank 0 prints one JSON line: whole-job samples/sec (sum over GPUs), bf16
+autocast compute, fp32 gradients compressed to 4 bits by cgx_hook.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "bert-large"])
+    p.add_argument("--batch", type=int, default=0, help="per-GPU batch")
+    p.add_argument("--bits", type=int, default=4)
+    p.add_argument("--bucket-size", type=int, default=1024)
+    p.add_argument("--backend", default="cgx", choices=["cgx", "nccl"])
+    p.add_argument("--no-compress", action="store_true",
+                   help="bits=32 (fp32 RCCL allreduce baseline)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    assert torch.cuda.is_available(), "bench.py requires a GPU"
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+    torch.manual_seed(1234 + rank)
+
+    bits = 32 if args.no_compress else args.bits
+    distributed = world > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        if args.backend == "cgx":
+            import torch_cgx_amd  # noqa: F401
+        dist.init_process_group(args.backend, rank=rank, world_size=world)
+
+    from torch_cgx_amd.models import resnet50, bert_large
+
+    if args.model == "resnet50":
+        batch = args.batch or 256
+        model = resnet50(num_classes=1000).to(device)
+        data = torch.randn(batch, 3, 224, 224, device=device)
+        target = torch.randint(0, 1000, (batch,), device=device)
+
+        def step_fn(m, opt):
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                loss = nn.functional.cross_entropy(m(data), target)
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            opt.step()
+            return loss
+    else:
+        batch = args.batch or 32
+        seq = 128
+        model = bert_large().to(device)
+        data = torch.randint(0, 30522, (batch, seq), device=device)
+        target = torch.randint(0, 30522, (batch, seq), device=device)
+
+        def step_fn(m, opt):
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                logits = m(data)
+                loss = nn.functional.cross_entropy(
+                    logits.view(-1, logits.size(-1)), target.view(-1))
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            opt.step()
+            return loss
+
+    if distributed:
+        model = nn.parallel.DistributedDataParallel(
+            model, device_ids=[local_rank], bucket_cap_mb=25)
+        if args.backend == "cgx":
+            import torch_cgx_amd
+            os.environ["CGX_COMPRESSION_QUANTIZATION_BITS"] = str(bits)
+            os.environ["CGX_COMPRESSION_BUCKET_SIZE"] = str(args.bucket_size)
+            state = torch_cgx_amd.CGXState(
+                None, layer_min_size=1024,
+                compression_params={"bits": bits,
+                                    "bucket_size": args.bucket_size})
+            model.register_comm_hook(state, torch_cgx_amd.cgx_hook)
+
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+
+    for _ in range(args.warmup):
+        step_fn(model, opt)
+
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step_fn(model, opt)
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    if distributed:  # max over ranks
+        e = torch.tensor([elapsed], device=device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = e.item()
+
+    n_gpus = world
+    samples_per_sec = batch * n_gpus * args.steps / elapsed
+    if rank == 0:
+        name = "ResNet-50" if args.model == "resnet50" else "BERT-large"
+        suffix = f"{bits}-bit" if bits <= 8 else "fp32"
+        result = {
+            "metric": f"DDP samples/sec {name} {suffix}",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch * n_gpus,
+                "input": "3x224x224" if args.model == "resnet50"
+                         else "seq128",
+                "parallelism": f"dp{n_gpus}",
+                "backend": args.backend,
+                "bits": bits,
+                "bucket_size": args.bucket_size,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -83,6 +83,36 @@ void py_dequantize(at::Tensor comp, at::Tensor out, int64_t bits,
       hb.cum[1], dt, (int)bits, stream.stream());
 }
 
+// Multi-source decode-and-sum: comp is [nsrc, stride] uint8; decodes each
+// row's stream and accumulates in T precision (the engine's round-1 path).
+void py_dequantize_multi(at::Tensor comp, at::Tensor out, int64_t bits,
+                         int64_t bucket_size, bool add) {
+  TORCH_CHECK(comp.is_cuda() && comp.dim() == 2 && comp.is_contiguous());
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous());
+  const DType dt = dtype_arg(out);
+  const int64_t n = out.numel();
+  if (n == 0) return;
+  struct Blob {
+    DequantDesc d;
+    int64_t cum[2];
+  } hb;
+  hb.d = DequantDesc{comp.data_ptr<uint8_t>(), out.data_ptr(), n,
+                     comp.stride(0), (int32_t)bucket_size,
+                     (int32_t)comp.size(0), add ? 1 : 0, 0};
+  hb.cum[0] = 0;
+  hb.cum[1] = (n + 7) / 8;
+  auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
+                           at::TensorOptions().dtype(at::kByte))
+                 .to(out.device());
+  auto stream =
+      c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(out.device().index());
+  const char* devp = static_cast<const char*>(dev.data_ptr());
+  launch_dequantize_batch(
+      reinterpret_cast<const DequantDesc*>(devp),
+      reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
+      hb.cum[1], dt, (int)bits, stream.stream());
+}
+
 int64_t py_buffer_size(int64_t n, at::ScalarType st, int64_t bits,
                        int64_t bucket_size) {
   DType dt = st == at::kFloat ? DType::F32
@@ -140,6 +170,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("seed") = 0);
   m.def("dequantize", &cgx::py_dequantize, py::arg("comp"), py::arg("out"),
         py::arg("bits"), py::arg("bucket_size"), py::arg("add") = false);
+  m.def("dequantize_multi", &cgx::py_dequantize_multi, py::arg("comp"),
+        py::arg("out"), py::arg("bits"), py::arg("bucket_size"),
+        py::arg("add") = false);
   m.def("buffer_size", &cgx::py_buffer_size);
   m.def("partition", &cgx::py_partition);
 }

@@ -1,2 +1,2 @@
-from .resnet import resnet50  # noqa: F401
-from .bert import bert_large  # noqa: F401
+from .resnet import resnet50, resnet18  # noqa: F401
+from .bert import bert_large, bert_tiny  # noqa: F401
