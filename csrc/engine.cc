@@ -107,6 +107,11 @@ EngineConfig EngineConfig::from_env() {
   c.default_bits = (int)env_int("CGX_COMPRESSION_QUANTIZATION_BITS", 32);
   c.default_bucket = (int)env_int("CGX_COMPRESSION_BUCKET_SIZE", 512);
   c.stochastic = env_int("CGX_STOCHASTIC_ROUNDING", 1) != 0;
+  const char* red = std::getenv("CGX_INNER_REDUCTION_TYPE");
+  if (!red || !*red) red = std::getenv("CGX_REDUCTION_TYPE");
+  c.ring = red && (std::strcmp(red, "Ring") == 0 ||
+                   std::strcmp(red, "ring") == 0 ||
+                   std::strcmp(red, "RING") == 0);
   return c;
 }
 
@@ -288,42 +293,48 @@ void Engine::run_dequant(const std::vector<Slice>& slices,
   }
 }
 
-void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
-                       ncclComm_t comm, hipStream_t stream,
-                       const EngineConfig& cfg) {
+Engine::ChunkPlan Engine::plan(const std::vector<LayerView>& views,
+                               DType dt) {
+  ChunkPlan pl;
   const int ws = size_;
   const int es = elem_size(dt);
-  int64_t n = 0;
   std::vector<int64_t> lnumels;
   lnumels.reserve(views.size());
   for (const auto& v : views) {
     lnumels.push_back(v.numel);
-    n += v.numel;
+    pl.n += v.numel;
   }
-  if (n == 0) return;
-
-  std::vector<int64_t> offs, szs;
-  partition(n, ws, lnumels, es, &offs, &szs);
-
-  // build per-rank slice lists + compressed chunk sizes
-  std::vector<std::vector<Slice>> rs(ws);
-  std::vector<int64_t> comp(ws, 0);
+  if (pl.n == 0) return pl;
+  partition(pl.n, ws, lnumels, es, &pl.offs, &pl.szs);
+  pl.rs.resize(ws);
+  pl.comp.assign(ws, 0);
   for (int r = 0; r < ws; r++) {
-    const int64_t start = offs[r], end = offs[r] + szs[r];
+    const int64_t start = pl.offs[r], end = pl.offs[r] + pl.szs[r];
     int64_t pos = 0, coff = 0;
     for (const auto& v : views) {
       const int64_t lo = std::max(pos, start);
       const int64_t hi = std::min(pos + v.numel, end);
       if (hi > lo) {
-        rs[r].push_back(Slice{v.data + (lo - pos) * es, hi - lo, v.bits,
-                              v.bucket_size, coff});
+        pl.rs[r].push_back(Slice{v.data + (lo - pos) * es, hi - lo, v.bits,
+                                 v.bucket_size, coff});
         coff += buffer_size(hi - lo, dt, v.bits, v.bucket_size);
       }
       pos += v.numel;
       if (pos >= end) break;
     }
-    comp[r] = coff;
+    pl.comp[r] = coff;
   }
+  return pl;
+}
+
+void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
+                       ncclComm_t comm, hipStream_t stream,
+                       const EngineConfig& cfg) {
+  const int ws = size_;
+  ChunkPlan pl = plan(views, dt);
+  if (pl.n == 0) return;
+  auto& rs = pl.rs;
+  auto& comp = pl.comp;
   const int64_t mycomp = comp[rank_];
 
   int64_t send1_total = 0;
@@ -411,6 +422,82 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
   }
 }
 
+void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
+                        ncclComm_t comm, hipStream_t stream,
+                        const EngineConfig& cfg) {
+  // Compressed ring allreduce (reference MPI_Allreduce_Ring semantics,
+  // ring.cc:139-226): ws-1 reduce-scatter steps with per-hop requantize of
+  // the running partial sum, then ws-1 allgather steps FORWARDING the
+  // once-quantized reduced segments, final batch decompress.
+  const int ws = size_;
+  ChunkPlan pl = plan(views, dt);
+  if (pl.n == 0) return;
+  auto& rs = pl.rs;
+  auto& comp = pl.comp;
+  const int next = (rank_ + 1) % ws;
+  const int prev = (rank_ + ws - 1) % ws;
+
+  int64_t seg_total = 0, maxc = 0;
+  std::vector<int64_t> seg_off(ws, 0);
+  for (int k = 0; k < ws; k++) {
+    seg_off[k] = seg_total;
+    seg_total += comp[k];
+    maxc = std::max(maxc, comp[k]);
+  }
+  uint8_t* base = staging(seg_total + 2 * maxc);
+  uint8_t* segs = base;
+  uint8_t* tmp_send = base + seg_total;
+  uint8_t* tmp_recv = tmp_send + maxc;
+
+  // reduce-scatter: accumulate into local chunks hop by hop
+  for (int s = 0; s < ws - 1; s++) {
+    const int send_c = (rank_ - s + ws) % ws;
+    const int recv_c = (rank_ - s - 1 + ws) % ws;
+    if (comp[send_c] > 0)
+      run_quantize(rs[send_c], tmp_send, dt, stream, cfg.stochastic);
+    CGX_NCCL_CHECK(ncclGroupStart());
+    if (comp[send_c] > 0)
+      CGX_NCCL_CHECK(
+          ncclSend(tmp_send, comp[send_c], ncclUint8, next, comm, stream));
+    if (comp[recv_c] > 0)
+      CGX_NCCL_CHECK(
+          ncclRecv(tmp_recv, comp[recv_c], ncclUint8, prev, comm, stream));
+    CGX_NCCL_CHECK(ncclGroupEnd());
+    if (comp[recv_c] > 0)
+      run_dequant(rs[recv_c], tmp_recv, 0, 1, /*add=*/true, dt, stream);
+  }
+
+  // own fully-reduced chunk after ws-1 hops
+  const int own = (rank_ + 1) % ws;
+  if (comp[own] > 0)
+    run_quantize(rs[own], segs + seg_off[own], dt, stream, cfg.stochastic);
+
+  // allgather: forward the quantized reduced segments around the ring
+  for (int s = 0; s < ws - 1; s++) {
+    const int send_c = (own - s + ws) % ws;
+    const int recv_c = (own - s - 1 + ws) % ws;
+    CGX_NCCL_CHECK(ncclGroupStart());
+    if (comp[send_c] > 0)
+      CGX_NCCL_CHECK(ncclSend(segs + seg_off[send_c], comp[send_c], ncclUint8,
+                              next, comm, stream));
+    if (comp[recv_c] > 0)
+      CGX_NCCL_CHECK(ncclRecv(segs + seg_off[recv_c], comp[recv_c], ncclUint8,
+                              prev, comm, stream));
+    CGX_NCCL_CHECK(ncclGroupEnd());
+  }
+
+  // final decode of every segment (own included, for bit-identical results)
+  std::vector<Slice> all;
+  for (int k = 0; k < ws; k++) {
+    for (const auto& sl : rs[k]) {
+      Slice t = sl;
+      t.comp_off += seg_off[k];
+      all.push_back(t);
+    }
+  }
+  run_dequant(all, segs, 0, 1, /*add=*/false, dt, stream);
+}
+
 void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
                        hipStream_t stream) {
   if (size_ <= 1) return;
@@ -467,9 +554,16 @@ void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   const int64_t fusion_elems = std::max<int64_t>(256, cfg.fusion_bytes / es);
   std::vector<LayerView> cur;
   int64_t cur_n = 0;
+  auto run_chunk = [&](const std::vector<LayerView>& vs) {
+    if (cfg.ring && size_ > 2) {
+      ring_chunk(vs, dt, comm, stream, cfg);
+    } else {
+      sra_chunk(vs, dt, comm, stream, cfg);
+    }
+  };
   auto flush = [&]() {
     if (!cur.empty()) {
-      sra_chunk(cur, dt, comm, stream, cfg);
+      run_chunk(cur);
       cur.clear();
       cur_n = 0;
     }
@@ -481,7 +575,7 @@ void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
         std::vector<LayerView> w{LayerView{v.data + o * es,
                                            std::min(fusion_elems, v.numel - o),
                                            v.bits, v.bucket_size}};
-        sra_chunk(w, dt, comm, stream, cfg);
+        run_chunk(w);
       }
     } else {
       if (cur_n + v.numel > fusion_elems) flush();

@@ -90,6 +90,7 @@ struct EngineConfig {
   int default_bits = 32;
   int default_bucket = 512;
   bool stochastic = true;
+  bool ring = false;  // CGX_INNER_REDUCTION_TYPE=Ring (default SRA)
   static EngineConfig from_env();  // re-read every bucket like the reference
 };
 
@@ -142,9 +143,20 @@ class Engine {
     int64_t comp_off;  // byte offset of this slice in the chunk's comp stream
   };
 
+  struct ChunkPlan {
+    std::vector<std::vector<Slice>> rs;  // per-rank slice lists
+    std::vector<int64_t> comp;           // per-rank compressed bytes
+    std::vector<int64_t> offs, szs;
+    int64_t n = 0;
+  };
+  ChunkPlan plan(const std::vector<LayerView>& views, DType dt);
+
   void sra_chunk(const std::vector<LayerView>& views, DType dt,
                  ncclComm_t comm, hipStream_t stream,
                  const EngineConfig& cfg);
+  void ring_chunk(const std::vector<LayerView>& views, DType dt,
+                  ncclComm_t comm, hipStream_t stream,
+                  const EngineConfig& cfg);
   uint8_t* staging(int64_t bytes);
 
   // Launch one quantize "job list", grouping slices by (bits, bucket%8==0).

@@ -56,7 +56,13 @@ def build(verbose: bool = True) -> str:
         src_abs = os.path.join(ROOT, src)
         if _mtime(obj) >= max(_mtime(src_abs), hdr_mtime):
             return obj
-        cmd = ["hipcc"] + flags + inc + ["-c", src_abs, "-o", obj]
+        extra = []
+        if src.endswith(".hip"):
+            # decode = round(unit*level) then round(min + product): FMA
+            # contraction would fold this into one rounding and break
+            # byte-parity with the golden wire-format model
+            extra = ["-ffp-contract=off"]
+        cmd = ["hipcc"] + flags + extra + inc + ["-c", src_abs, "-o", obj]
         if verbose:
             print("[cgx build]", src, flush=True)
         r = subprocess.run(cmd, capture_output=True, text=True)
