@@ -160,16 +160,43 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
       (t.scalar_type() == at::kFloat || t.scalar_type() == at::kHalf ||
        t.scalar_type() == at::kBFloat16);
   auto op = opts.reduceOp;
+  if (compressible) {
+    // pipelined path: engine fans work over quantize/comm/deq streams and
+    // returns the stream carrying the final op; the Work's end event and
+    // future live there so the NEXT bucket's quantize can overlap this
+    // bucket's xGMI traffic.
+    std::lock_guard<std::mutex> lock(mu_);
+    lazyInit(t.device());
+    c10::hip::HIPGuardMasqueradingAsCUDA dguard(device_index_);
+    auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_index_);
+    CGX_HIP_CHECK(hipEventRecord(start_ev_, cur.stream()));
+    CGX_HIP_CHECK(hipStreamWaitEvent(stream_->stream(), start_ev_, 0));
+    hipStream_t fin = stream_->stream();
+    if (size_ > 1) fin = engine_->allreduce(t, comm_, stream_->stream());
+    auto fin_masq = c10::hip::getStreamFromExternalMasqueradingAsCUDA(
+        fin, device_index_);
+    c10::hip::HIPStreamGuardMasqueradingAsCUDA sguard(fin_masq.unwrap());
+    for (hipStream_t s : {stream_->stream(), engine_ ? engine_->comm_stream()
+                                                     : nullptr,
+                          engine_ ? engine_->deq_stream() : nullptr}) {
+      if (!s) continue;
+      c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
+          recordStreamMasqueradingAsCUDA(
+              t.storage().data_ptr(),
+              c10::hip::getStreamFromExternalMasqueradingAsCUDA(
+                  s, device_index_));
+    }
+    auto work = c10::make_intrusive<WorkCGX>(rank_, c10d::OpType::ALLREDUCE,
+                                             t.device(), tensors);
+    work->recordEnd(fin_masq);
+    return work;
+  }
   return collective(tensors, t.device(), c10d::OpType::ALLREDUCE,
-                    [this, t, compressible, op](hipStream_t s) {
+                    [this, t, op](hipStream_t s) {
                       if (size_ == 1) return;
-                      if (compressible) {
-                        engine_->allreduce(t, comm_, s);
-                      } else {
-                        CGX_NCCL_CHECK(ncclAllReduce(
-                            t.data_ptr(), t.data_ptr(), t.numel(),
-                            nccl_dtype(t), to_nccl_op(op), comm_, s));
-                      }
+                      CGX_NCCL_CHECK(ncclAllReduce(
+                          t.data_ptr(), t.data_ptr(), t.numel(),
+                          nccl_dtype(t), to_nccl_op(op), comm_, s));
                     });
 }
 
@@ -516,6 +543,10 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::barrier(
   if (comm_) {
     std::lock_guard<std::mutex> lock(mu_);
     CGX_HIP_CHECK(hipStreamSynchronize(stream_->stream()));
+    if (engine_) {
+      CGX_HIP_CHECK(hipStreamSynchronize(engine_->comm_stream()));
+      CGX_HIP_CHECK(hipStreamSynchronize(engine_->deq_stream()));
+    }
   }
   if (cpu_) {
     c10d::BarrierOptions o = opts;

@@ -1,6 +1,8 @@
 // Implementation of the compression engine (see engine.h).
 #include "engine.h"
 
+#include <ATen/hip/impl/HIPCachingAllocatorMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <c10/util/Exception.h>
 
 #include <algorithm>
@@ -187,6 +189,38 @@ ncclDataType_t nccl_dtype(const at::Tensor& t) {
 // ---------------------------------------------------------------------------
 Engine::Engine(int rank, int size) : rank_(rank), size_(size) {
   seed_ = 0x9E3779B97F4A7C15ull ^ (0xD1B54A32D192ED03ull * (uint64_t)(rank + 1));
+  // constructed under the backend's device guard (lazyInit)
+  int least = 0, greatest = 0;
+  CGX_HIP_CHECK(hipDeviceGetStreamPriorityRange(&least, &greatest));
+  CGX_HIP_CHECK(hipStreamCreateWithPriority(&comm_stream_,
+                                            hipStreamNonBlocking, greatest));
+  CGX_HIP_CHECK(hipStreamCreateWithPriority(&deq_stream_,
+                                            hipStreamNonBlocking, greatest));
+  for (auto& e : evs_)
+    CGX_HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+  for (auto& s : slots_)
+    CGX_HIP_CHECK(hipEventCreateWithFlags(&s.done_ev, hipEventDisableTiming));
+}
+
+Engine::~Engine() {
+  for (auto& s : slots_)
+    if (s.done_ev) (void)hipEventDestroy(s.done_ev);
+  for (auto& e : evs_)
+    if (e) (void)hipEventDestroy(e);
+  if (comm_stream_) (void)hipStreamDestroy(comm_stream_);
+  if (deq_stream_) (void)hipStreamDestroy(deq_stream_);
+}
+
+hipEvent_t Engine::next_ev() {
+  hipEvent_t e = evs_[ev_cur_];
+  ev_cur_ = (ev_cur_ + 1) % kEvents;
+  return e;
+}
+
+void Engine::chain(hipStream_t from, hipStream_t to) {
+  hipEvent_t e = next_ev();
+  CGX_HIP_CHECK(hipEventRecord(e, from));
+  CGX_HIP_CHECK(hipStreamWaitEvent(to, e, 0));
 }
 
 void Engine::partition(int64_t num_elements, int ws,
@@ -233,6 +267,31 @@ uint8_t* Engine::staging(int64_t bytes) {
                                     .device(at::kCUDA));
   }
   return staging_.data_ptr<uint8_t>();
+}
+
+uint8_t* Engine::slot_bytes(StagingSlot& slot, int64_t bytes) {
+  if (bytes <= 0) bytes = 1;
+  if (!slot.buf.defined() || slot.buf.numel() < bytes) {
+    if (slot.buf.defined()) {
+      // the old buffer may still be referenced by in-flight comm/deq work;
+      // let the caching allocator defer reuse until those streams pass
+      auto rec = [&](hipStream_t s) {
+        c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
+            recordStreamMasqueradingAsCUDA(
+                slot.buf.storage().data_ptr(),
+                c10::hip::getStreamFromExternalMasqueradingAsCUDA(
+                    s, slot.buf.device().index()));
+      };
+      rec(comm_stream_);
+      rec(deq_stream_);
+    }
+    int64_t cap = 1 << 20;
+    while (cap < bytes) cap *= 2;
+    slot.buf = at::empty({cap}, at::TensorOptions()
+                                    .dtype(at::kByte)
+                                    .device(at::kCUDA));
+  }
+  return slot.buf.data_ptr<uint8_t>();
 }
 
 void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
@@ -328,7 +387,7 @@ Engine::ChunkPlan Engine::plan(const std::vector<LayerView>& views,
 }
 
 void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
-                       ncclComm_t comm, hipStream_t stream,
+                       ncclComm_t comm, hipStream_t qs,
                        const EngineConfig& cfg) {
   const int ws = size_;
   ChunkPlan pl = plan(views, dt);
@@ -341,7 +400,15 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
   for (int p = 0; p < ws; p++)
     if (p != rank_) send1_total += comp[p];
   const int64_t recv1_total = (int64_t)(ws - 1) * mycomp;
-  uint8_t* base = staging(send1_total + recv1_total + mycomp + send1_total);
+
+  StagingSlot& sslot = slots_[slot_cur_];
+  slot_cur_ ^= 1;
+  if (sslot.recorded) {
+    // don't overwrite staging another chunk's comm/decode still reads
+    CGX_HIP_CHECK(hipStreamWaitEvent(qs, sslot.done_ev, 0));
+  }
+  uint8_t* base =
+      slot_bytes(sslot, send1_total + recv1_total + mycomp + send1_total);
   uint8_t* send1 = base;
   uint8_t* recv1 = send1 + send1_total;
   uint8_t* send2 = recv1 + recv1_total;
@@ -358,68 +425,76 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
   }
   auto slot = [&](int p) { return p < rank_ ? p : p - 1; };
 
-  // round 1: quantize my copy of every peer's chunk
+  // round 1 quantize of every peer chunk, on the quantize stream
   {
     std::vector<Slice> all;
     for (int p = 0; p < ws; p++) {
       if (p == rank_) continue;
-      for (const auto& s : rs[p]) {
-        Slice t = s;
+      for (const auto& sl : rs[p]) {
+        Slice t = sl;
         t.comp_off += peer_off[p];
         all.push_back(t);
       }
     }
-    run_quantize(all, send1, dt, stream, cfg.stochastic);
+    run_quantize(all, send1, dt, qs, cfg.stochastic);
   }
 
-  // round 1: exchange (grouped p2p drives all xGMI links concurrently)
+  // round 1 exchange on the comm stream (grouped p2p drives all xGMI links)
+  chain(qs, comm_stream_);
   CGX_NCCL_CHECK(ncclGroupStart());
   for (int p = 0; p < ws; p++) {
     if (p == rank_) continue;
     if (comp[p] > 0)
       CGX_NCCL_CHECK(ncclSend(send1 + peer_off[p], comp[p], ncclUint8, p,
-                              comm, stream));
+                              comm, comm_stream_));
     if (mycomp > 0)
       CGX_NCCL_CHECK(ncclRecv(recv1 + (int64_t)slot(p) * mycomp, mycomp,
-                              ncclUint8, p, comm, stream));
+                              ncclUint8, p, comm, comm_stream_));
   }
   CGX_NCCL_CHECK(ncclGroupEnd());
 
-  // reduce: my raw chunk += sum of (ws-1) decoded streams (peer-ascending)
+  // decode-accumulate + self-quantize on the deq stream
+  chain(comm_stream_, deq_stream_);
   if (mycomp > 0) {
-    run_dequant(rs[rank_], recv1, mycomp, ws - 1, /*add=*/true, dt, stream);
+    run_dequant(rs[rank_], recv1, mycomp, ws - 1, /*add=*/true, dt,
+                deq_stream_);
     // self-quantize the reduced chunk; the same bytes go to every peer and
     // through my own decode so all ranks end bit-identical
-    run_quantize(rs[rank_], send2, dt, stream, cfg.stochastic);
+    run_quantize(rs[rank_], send2, dt, deq_stream_, cfg.stochastic);
   }
 
-  // round 2: allgather of reduced chunks
+  // round 2 exchange on the comm stream
+  chain(deq_stream_, comm_stream_);
   CGX_NCCL_CHECK(ncclGroupStart());
   for (int p = 0; p < ws; p++) {
     if (p == rank_) continue;
     if (mycomp > 0)
-      CGX_NCCL_CHECK(ncclSend(send2, mycomp, ncclUint8, p, comm, stream));
+      CGX_NCCL_CHECK(
+          ncclSend(send2, mycomp, ncclUint8, p, comm, comm_stream_));
     if (comp[p] > 0)
       CGX_NCCL_CHECK(ncclRecv(recv2 + peer_off[p], comp[p], ncclUint8, p,
-                              comm, stream));
+                              comm, comm_stream_));
   }
   CGX_NCCL_CHECK(ncclGroupEnd());
 
   // final decode: own chunk from send2, peers' chunks from recv2
+  chain(comm_stream_, deq_stream_);
   if (mycomp > 0)
-    run_dequant(rs[rank_], send2, 0, 1, /*add=*/false, dt, stream);
+    run_dequant(rs[rank_], send2, 0, 1, /*add=*/false, dt, deq_stream_);
   {
     std::vector<Slice> all;
     for (int p = 0; p < ws; p++) {
       if (p == rank_) continue;
-      for (const auto& s : rs[p]) {
-        Slice t = s;
+      for (const auto& sl : rs[p]) {
+        Slice t = sl;
         t.comp_off += peer_off[p];
         all.push_back(t);
       }
     }
-    run_dequant(all, recv2, 0, 1, /*add=*/false, dt, stream);
+    run_dequant(all, recv2, 0, 1, /*add=*/false, dt, deq_stream_);
   }
+  CGX_HIP_CHECK(hipEventRecord(sslot.done_ev, deq_stream_));
+  sslot.recorded = true;
 }
 
 void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
@@ -498,9 +573,9 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
   run_dequant(all, segs, 0, 1, /*add=*/false, dt, stream);
 }
 
-void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
-                       hipStream_t stream) {
-  if (size_ <= 1) return;
+hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
+                              hipStream_t qs) {
+  if (size_ <= 1) return qs;
   TORCH_CHECK(bucket.is_contiguous(), "cgx: bucket must be contiguous");
   const EngineConfig cfg = EngineConfig::from_env();
   const DType dt = dtype_of(bucket);
@@ -542,10 +617,11 @@ void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
 
   if (!uncomp.empty()) {
     const ncclDataType_t ndt = nccl_dtype(bucket);
+    chain(qs, comm_stream_);
     CGX_NCCL_CHECK(ncclGroupStart());
     for (auto& [ptr, cnt] : uncomp) {
       CGX_NCCL_CHECK(
-          ncclAllReduce(ptr, ptr, cnt, ndt, ncclSum, comm, stream));
+          ncclAllReduce(ptr, ptr, cnt, ndt, ncclSum, comm, comm_stream_));
     }
     CGX_NCCL_CHECK(ncclGroupEnd());
   }
@@ -554,11 +630,14 @@ void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   const int64_t fusion_elems = std::max<int64_t>(256, cfg.fusion_bytes / es);
   std::vector<LayerView> cur;
   int64_t cur_n = 0;
+  bool any_comp = false;
   auto run_chunk = [&](const std::vector<LayerView>& vs) {
+    any_comp = true;
     if (cfg.ring && size_ > 2) {
-      ring_chunk(vs, dt, comm, stream, cfg);
+      ring_chunk(vs, dt, comm, qs, cfg);  // ring is hop-serial: one stream
+      chain(qs, deq_stream_);             // keep completion on deq stream
     } else {
-      sra_chunk(vs, dt, comm, stream, cfg);
+      sra_chunk(vs, dt, comm, qs, cfg);
     }
   };
   auto flush = [&]() {
@@ -584,6 +663,16 @@ void Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
     }
   }
   flush();
+
+  // completion stream: all compressed chunks end on deq_stream_; the
+  // uncompressed group ran (first) on comm_stream_, which every later chunk
+  // already chained through -- but cover the uncomp-only case explicitly.
+  if (any_comp) {
+    if (!uncomp.empty() && comp_views.empty()) chain(comm_stream_, deq_stream_);
+    return deq_stream_;
+  }
+  if (!uncomp.empty()) return comm_stream_;
+  return qs;
 }
 
 }  // namespace cgx

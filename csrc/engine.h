@@ -121,11 +121,22 @@ class DescRing {
 class Engine {
  public:
   Engine(int rank, int size);
+  ~Engine();
 
-  // SUM-allreduce of a flat fp32/fp16/bf16 CUDA tensor on `stream`:
-  // registered (or default-config) layers are quantized and reduced via SRA
-  // over RCCL; non-compressible layers go through fused ncclAllReduce.
-  void allreduce(at::Tensor bucket, ncclComm_t comm, hipStream_t stream);
+  // SUM-allreduce of a flat fp32/fp16/bf16 CUDA tensor: registered (or
+  // default-config) layers are quantized and reduced via SRA over RCCL;
+  // non-compressible layers go through fused ncclAllReduce.
+  //
+  // Three-stream pipeline: quantize runs on the caller's side stream `qs`,
+  // p2p/collective traffic on comm_stream_, decode(+re-quantize) on
+  // deq_stream_, chained by events.  Because `qs` never waits on the other
+  // two, the NEXT bucket's quantize overlaps this bucket's xGMI traffic
+  // (the reference serialized everything on one side stream).  Returns the
+  // stream carrying the final operation (record the Work end event there).
+  hipStream_t allreduce(at::Tensor bucket, ncclComm_t comm, hipStream_t qs);
+
+  hipStream_t comm_stream() const { return comm_stream_; }
+  hipStream_t deq_stream() const { return deq_stream_; }
 
   // Mirror of the partition walk (reference Quantizer::GetSizesAndOffsets,
   // compressor.cc:265-299); exposed for tests via bindings.
@@ -143,6 +154,14 @@ class Engine {
     int64_t comp_off;  // byte offset of this slice in the chunk's comp stream
   };
 
+  // Double-buffered staging so chunk c+1's quantize (on qs) can start while
+  // chunk c's traffic/decode (on comm/deq streams) still reads its slot.
+  struct StagingSlot {
+    at::Tensor buf;
+    hipEvent_t done_ev = nullptr;  // recorded on deq stream at chunk end
+    bool recorded = false;
+  };
+
   struct ChunkPlan {
     std::vector<std::vector<Slice>> rs;  // per-rank slice lists
     std::vector<int64_t> comp;           // per-rank compressed bytes
@@ -152,12 +171,14 @@ class Engine {
   ChunkPlan plan(const std::vector<LayerView>& views, DType dt);
 
   void sra_chunk(const std::vector<LayerView>& views, DType dt,
-                 ncclComm_t comm, hipStream_t stream,
-                 const EngineConfig& cfg);
+                 ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg);
   void ring_chunk(const std::vector<LayerView>& views, DType dt,
                   ncclComm_t comm, hipStream_t stream,
                   const EngineConfig& cfg);
-  uint8_t* staging(int64_t bytes);
+  uint8_t* staging(int64_t bytes);                       // single-stream path
+  uint8_t* slot_bytes(StagingSlot& slot, int64_t bytes); // pipelined path
+  void chain(hipStream_t from, hipStream_t to);  // event: `to` waits `from`
+  hipEvent_t next_ev();
 
   // Launch one quantize "job list", grouping slices by (bits, bucket%8==0).
   void run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
@@ -168,6 +189,13 @@ class Engine {
 
   int rank_, size_;
   at::Tensor staging_;
+  StagingSlot slots_[2];
+  int slot_cur_ = 0;
+  hipStream_t comm_stream_ = nullptr;
+  hipStream_t deq_stream_ = nullptr;
+  static constexpr int kEvents = 16;
+  hipEvent_t evs_[kEvents] = {};
+  int ev_cur_ = 0;
   DescRing ring_;
   uint64_t seed_;
 };
