@@ -30,12 +30,18 @@ constexpr int kThreads = 256;  // 4 waves
 constexpr int kWave = 64;
 constexpr int kMaxBlocks = 4096;  // 256 CU * 8 blocks/CU * 2
 
-__device__ __forceinline__ float randf(uint64_t seed, uint64_t key) {
+// One splitmix64-style hash per 8-value pack; each element draws 8 bits of
+// uniform randomness ([0,1) in steps of 1/256 -- rounding bias <= 1/512 of a
+// quantization unit, negligible vs. the unit-sized quantization error).
+__device__ __forceinline__ uint64_t rand_pack(uint64_t seed, uint64_t key) {
   uint64_t z = seed + key * 0x9E3779B97F4A7C15ull;
   z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
   z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
-  z ^= z >> 31;
-  return static_cast<float>(z >> 40) * (1.0f / 16777216.0f);  // [0, 1)
+  return z ^ (z >> 31);
+}
+
+__device__ __forceinline__ float rand_lane(uint64_t pack_rand, int j) {
+  return static_cast<float>((pack_rand >> (8 * j)) & 0xFF) * (1.0f / 256.0f);
 }
 
 template <typename To, typename From>
@@ -258,6 +264,7 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
       const int64_t num_char = (d.n * BITS + 7) >> 3;
       const int64_t gbase = bstart >> 3;
       const bool live = unitf >= kEps;
+      const float rinv = 1.0f / unitf;  // hoisted: fp32 div is 1/4 VALU rate
       for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
         uint32_t r[8];
         const int m = min(8, cur - g * 8);
@@ -272,14 +279,13 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
         }
         uint64_t value = 0;
         if (live) {
+          const uint64_t pr =
+              stochastic ? rand_pack(seed, (static_cast<uint64_t>(lo) << 44) |
+                                               static_cast<uint64_t>(gbase + g))
+                         : 0;
           for (int j = 0; j < m; j++) {
-            const int64_t eidx = bstart + g * 8 + j;
-            const float rnd =
-                stochastic
-                    ? randf(seed, (static_cast<uint64_t>(lo) << 44) |
-                                      static_cast<uint64_t>(eidx))
-                    : 0.5f;
-            const float dd = (raw2f<T>(r[j]) - minf) / unitf + rnd;
+            const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
+            const float dd = (raw2f<T>(r[j]) - minf) * rinv + rnd;
             const uint32_t level =
                 static_cast<uint32_t>(fminf(floorf(dd), divisor));
             value |= static_cast<uint64_t>(level & ((1u << BITS) - 1))
@@ -315,20 +321,26 @@ __global__ __launch_bounds__(kThreads) void k_pack_generic(
         reinterpret_cast<uint8_t*>(d.out) + 2 * sizeof(R) * nb_slice;
     const int64_t num_char = (d.n * BITS + 7) >> 3;
     const R* in = reinterpret_cast<const R*>(d.in);
+    const bool small = d.n < (int64_t(1) << 31);
     for (int64_t g = t0; g < ngroups; g += stride) {
       const int m = static_cast<int>(min(static_cast<int64_t>(8), d.n - g * 8));
       uint64_t value = 0;
+      const uint64_t pr =
+          stochastic ? rand_pack(seed, (static_cast<uint64_t>(s) << 44) |
+                                           static_cast<uint64_t>(g))
+                     : 0;
       for (int j = 0; j < m; j++) {
         const int64_t eidx = g * 8 + j;
-        const int64_t bk = eidx / d.bucket;
+        const int64_t bk =
+            small ? static_cast<int64_t>(static_cast<uint32_t>(eidx) /
+                                         static_cast<uint32_t>(d.bucket))
+                  : eidx / d.bucket;
         const float unitf = raw2f<T>(meta[2 * bk]);
         if (unitf < kEps) continue;
         const float minf = raw2f<T>(meta[2 * bk + 1]);
-        const float rnd =
-            stochastic ? randf(seed, (static_cast<uint64_t>(s) << 44) |
-                                         static_cast<uint64_t>(eidx))
-                       : 0.5f;
-        const float dd = (raw2f<T>(in[eidx]) - minf) / unitf + rnd;
+        const float rinv = 1.0f / unitf;
+        const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
+        const float dd = (raw2f<T>(in[eidx]) - minf) * rinv + rnd;
         const uint32_t level = static_cast<uint32_t>(fminf(floorf(dd), divisor));
         value |= static_cast<uint64_t>(level & ((1u << BITS) - 1)) << (j * BITS);
       }
@@ -373,6 +385,7 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
     const bool al16 =
         m == 8 && (reinterpret_cast<uintptr_t>(outp) & 15) == 0;
     const bool one_bucket = (d.bucket & 7) == 0;
+    const bool small = d.n < (int64_t(1) << 31);
 
     uint32_t v[8];
     bool have = d.add != 0;
@@ -388,9 +401,20 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       const uint8_t* src = d.in + s * d.src_stride;
       const R* meta = reinterpret_cast<const R*>(src);
       const uint64_t value = load_bytes(src + meta_bytes + gb, nbytes);
-      int64_t bk0 = one_bucket ? (g * 8) / d.bucket : 0;
+      const int64_t bk0 =
+          one_bucket
+              ? (small ? static_cast<int64_t>(
+                             static_cast<uint32_t>(g * 8) /
+                             static_cast<uint32_t>(d.bucket))
+                       : (g * 8) / d.bucket)
+              : 0;
       for (int j = 0; j < m; j++) {
-        const int64_t bk = one_bucket ? bk0 : (g * 8 + j) / d.bucket;
+        const int64_t bk =
+            one_bucket ? bk0
+                       : (small ? static_cast<int64_t>(
+                                      static_cast<uint32_t>(g * 8 + j) /
+                                      static_cast<uint32_t>(d.bucket))
+                                : (g * 8 + j) / d.bucket);
         const uint32_t lvl =
             static_cast<uint32_t>((value >> (j * BITS)) & ((1u << BITS) - 1));
         const float unitf = raw2f<T>(meta[2 * bk]);

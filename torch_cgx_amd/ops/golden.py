@@ -21,7 +21,8 @@ For a 1-D buffer of ``n`` elements of dtype ``T`` (fp32 / fp16 / bf16),
   consecutive values is encoded into a 64-bit word
   ``value = sum(level_j << (j * q))`` and written as its low
   ``q`` bytes, little-endian; total packed bytes ``num_char = ceil(n*q/8)``.
-* encode: ``level = min(floor((x - min)/unit + rand), 2**q - 1)`` in fp32
+* encode: ``level = min(floor((x - min) * (1/unit) + rand), 2**q - 1)``
+  in fp32, with the per-bucket reciprocal ``1/unit`` rounded to fp32 once
   (``level = 0`` when ``unit < EPS``); deterministic rounding uses
   ``rand = 0.5``.
 * decode: ``min + unit * level`` computed in dtype T.
@@ -103,8 +104,13 @@ def encode_levels(x: torch.Tensor, meta: torch.Tensor, bits: int,
     xf = x.float().numpy()
     if isinstance(rand, torch.Tensor):
         rand = rand.float().numpy()
+    # encode via the fp32-rounded per-bucket reciprocal (the kernels hoist
+    # 1/unit out of the element loop; fp division per element is ~4x VALU
+    # cost on CDNA4).  All fp32 ops single-rounded -> bitwise kernel parity.
     with np.errstate(divide="ignore", invalid="ignore"):
-        d = (xf - bmin[idx]) / unit[idx] + rand
+        inv_unit = (np.float32(1.0) / unit).astype(np.float32)
+        rnd = np.float32(rand) if np.isscalar(rand) else rand.astype(np.float32)
+        d = (xf - bmin[idx]).astype(np.float32) * inv_unit[idx] + rnd
     level = np.minimum(np.floor(d), float((1 << bits) - 1))
     level = np.where(unit[idx] < EPS, 0.0, level)
     return level.astype(np.uint8)
