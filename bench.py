@@ -36,6 +36,9 @@ def parse_args():
     p.add_argument("--backend", default="cgx", choices=["cgx", "nccl"])
     p.add_argument("--no-compress", action="store_true",
                    help="bits=32 (fp32 RCCL allreduce baseline)")
+    p.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
+                   help="cpu: harness validation only (gloo delegation)")
+    p.add_argument("--channels-last", type=int, default=1)
     return p.parse_args()
 
 
@@ -44,10 +47,19 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    assert torch.cuda.is_available(), "bench.py requires a GPU"
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    use_cuda = args.device == "cuda"
+    if use_cuda:
+        assert torch.cuda.is_available(), "bench.py requires a GPU"
+        torch.cuda.set_device(local_rank)
+        torch.backends.cudnn.benchmark = True
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
     torch.manual_seed(1234 + rank)
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
 
     bits = 32 if args.no_compress else args.bits
     distributed = world > 1
@@ -61,27 +73,38 @@ def main():
     from torch_cgx_amd.models import resnet50, bert_large
 
     if args.model == "resnet50":
-        batch = args.batch or 256
+        batch = args.batch or (256 if use_cuda else 4)
         model = resnet50(num_classes=1000).to(device)
         data = torch.randn(batch, 3, 224, 224, device=device)
         target = torch.randint(0, 1000, (batch,), device=device)
+        if use_cuda and args.channels_last:
+            model = model.to(memory_format=torch.channels_last)
+            data = data.to(memory_format=torch.channels_last)
 
         def step_fn(m, opt):
-            with torch.autocast("cuda", dtype=torch.bfloat16):
+            if use_cuda:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    loss = nn.functional.cross_entropy(m(data), target)
+            else:
                 loss = nn.functional.cross_entropy(m(data), target)
             opt.zero_grad(set_to_none=True)
             loss.backward()
             opt.step()
             return loss
     else:
-        batch = args.batch or 32
+        batch = args.batch or (32 if use_cuda else 2)
         seq = 128
         model = bert_large().to(device)
         data = torch.randint(0, 30522, (batch, seq), device=device)
         target = torch.randint(0, 30522, (batch, seq), device=device)
 
         def step_fn(m, opt):
-            with torch.autocast("cuda", dtype=torch.bfloat16):
+            if use_cuda:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    logits = m(data)
+                    loss = nn.functional.cross_entropy(
+                        logits.view(-1, logits.size(-1)), target.view(-1))
+            else:
                 logits = m(data)
                 loss = nn.functional.cross_entropy(
                     logits.view(-1, logits.size(-1)), target.view(-1))
@@ -92,7 +115,8 @@ def main():
 
     if distributed:
         model = nn.parallel.DistributedDataParallel(
-            model, device_ids=[local_rank], bucket_cap_mb=25)
+            model, device_ids=[local_rank] if use_cuda else None,
+            bucket_cap_mb=25)
         if args.backend == "cgx":
             import torch_cgx_amd
             os.environ["CGX_COMPRESSION_QUANTIZATION_BITS"] = str(bits)
@@ -111,13 +135,13 @@ def main():
 
     if distributed:
         dist.barrier()
-    torch.cuda.synchronize()
+    sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step_fn(model, opt)
     if distributed:
         dist.barrier()
-    torch.cuda.synchronize()
+    sync()
     elapsed = time.perf_counter() - t0
 
     if distributed:  # max over ranks
@@ -141,7 +165,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if use_cuda else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -109,6 +109,9 @@ EngineConfig EngineConfig::from_env() {
   c.default_bits = (int)env_int("CGX_COMPRESSION_QUANTIZATION_BITS", 32);
   c.default_bucket = (int)env_int("CGX_COMPRESSION_BUCKET_SIZE", 512);
   c.stochastic = env_int("CGX_STOCHASTIC_ROUNDING", 1) != 0;
+  const char* fr = std::getenv("CGX_COMPRESSION_FAKE_RATIO");
+  if (fr && *fr) c.fake_ratio = std::atof(fr);
+  if (!(c.fake_ratio > 0.0 && c.fake_ratio <= 1.0)) c.fake_ratio = 1.0;
   const char* red = std::getenv("CGX_INNER_REDUCTION_TYPE");
   if (!red || !*red) red = std::getenv("CGX_REDUCTION_TYPE");
   c.ring = red && (std::strcmp(red, "Ring") == 0 ||
@@ -631,8 +634,27 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   std::vector<LayerView> cur;
   int64_t cur_n = 0;
   bool any_comp = false;
-  auto run_chunk = [&](const std::vector<LayerView>& vs) {
+  auto run_chunk = [&](const std::vector<LayerView>& vs_in) {
     any_comp = true;
+    // CGX_COMPRESSION_FAKE_RATIO < 1: reduce only a fraction of each chunk
+    // (bandwidth experiments; intentionally lossy -- reference
+    // mpi_allreduce_operations.cc:143-144)
+    std::vector<LayerView> trimmed;
+    const std::vector<LayerView>* vsp = &vs_in;
+    if (cfg.fake_ratio < 1.0) {
+      int64_t total = 0;
+      for (const auto& v : vs_in) total += v.numel;
+      int64_t keep = std::max<int64_t>(16, (int64_t)(total * cfg.fake_ratio));
+      for (const auto& v : vs_in) {
+        if (keep <= 0) break;
+        LayerView w = v;
+        w.numel = std::min(v.numel, keep);
+        keep -= w.numel;
+        trimmed.push_back(w);
+      }
+      vsp = &trimmed;
+    }
+    const std::vector<LayerView>& vs = *vsp;
     if (cfg.ring && size_ > 2) {
       ring_chunk(vs, dt, comm, qs, cfg);  // ring is hop-serial: one stream
       chain(qs, deq_stream_);             // keep completion on deq stream

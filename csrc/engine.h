@@ -91,6 +91,7 @@ struct EngineConfig {
   int default_bucket = 512;
   bool stochastic = true;
   bool ring = false;  // CGX_INNER_REDUCTION_TYPE=Ring (default SRA)
+  double fake_ratio = 1.0;  // CGX_COMPRESSION_FAKE_RATIO (bandwidth expt)
   static EngineConfig from_env();  // re-read every bucket like the reference
 };
 
