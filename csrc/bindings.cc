@@ -22,13 +22,14 @@ DType dtype_arg(const at::Tensor& t) { return dtype_of(t); }
 // Compress a 1-D CUDA tensor; returns the uint8 compressed buffer
 // (zero-filled alignment padding so byte comparisons are well-defined).
 at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
-                       bool stochastic, int64_t seed) {
+                       bool stochastic, int64_t seed, bool skip_incomplete) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "quantize: CUDA contiguous");
   TORCH_CHECK(bits >= 1 && bits <= 8, "quantize: bits in [1,8]");
   TORCH_CHECK(bucket_size >= 1, "quantize: bucket_size >= 1");
   const DType dt = dtype_arg(x);
   const int64_t n = x.numel();
-  const int64_t bytes = buffer_size(n, dt, (int)bits, (int)bucket_size);
+  const int64_t bytes =
+      buffer_size(n, dt, (int)bits, (int)bucket_size, skip_incomplete);
   auto out = at::zeros({std::max<int64_t>(bytes, 1)},
                        at::TensorOptions().dtype(at::kByte).device(x.device()));
   if (n == 0) return out;
@@ -38,9 +39,11 @@ at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
     int64_t cum[2];
   } hb;
   hb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), n,
-                   (int32_t)bucket_size, 0};
+                   (int32_t)bucket_size,
+                   skip_incomplete ? kFlagSkipIncomplete : 0};
   hb.cum[0] = 0;
-  hb.cum[1] = (n + bucket_size - 1) / bucket_size;
+  hb.cum[1] = skip_incomplete ? n / bucket_size
+                              : (n + bucket_size - 1) / bucket_size;
   auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
                            at::TensorOptions().dtype(at::kByte))
                  .to(x.device());
@@ -51,26 +54,30 @@ at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
       reinterpret_cast<const QuantDesc*>(devp),
       reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
       hb.cum[1], dt, (int)bits, (uint64_t)seed, stochastic, stream.stream(),
-      bucket_size % 8 == 0);
+      bucket_size % 8 == 0,
+      skip_incomplete && (n % bucket_size) != 0);
   return out;
 }
 
 // Decompress `comp` into `out` (1-D CUDA tensor of n elements); add=True
 // accumulates in T precision.
 void py_dequantize(at::Tensor comp, at::Tensor out, int64_t bits,
-                   int64_t bucket_size, bool add) {
+                   int64_t bucket_size, bool add, bool skip_incomplete) {
   TORCH_CHECK(comp.is_cuda() && out.is_cuda() && out.is_contiguous());
   const DType dt = dtype_arg(out);
   const int64_t n = out.numel();
   if (n == 0) return;
+  const int64_t nq =
+      skip_incomplete ? n / bucket_size * bucket_size : n;
   struct Blob {
     DequantDesc d;
     int64_t cum[2];
   } hb;
   hb.d = DequantDesc{comp.data_ptr<uint8_t>(), out.data_ptr(), n, 0,
-                     (int32_t)bucket_size, 1, add ? 1 : 0, 0};
+                     (int32_t)bucket_size, 1, add ? 1 : 0,
+                     skip_incomplete ? kFlagSkipIncomplete : 0};
   hb.cum[0] = 0;
-  hb.cum[1] = (n + 7) / 8;
+  hb.cum[1] = (nq + 7) / 8;
   auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
                            at::TensorOptions().dtype(at::kByte))
                  .to(out.device());
@@ -80,7 +87,8 @@ void py_dequantize(at::Tensor comp, at::Tensor out, int64_t bits,
   launch_dequantize_batch(
       reinterpret_cast<const DequantDesc*>(devp),
       reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
-      hb.cum[1], dt, (int)bits, stream.stream());
+      hb.cum[1], dt, (int)bits, stream.stream(),
+      skip_incomplete && (n % bucket_size) != 0);
 }
 
 // Multi-source decode-and-sum: comp is [nsrc, stride] uint8; decodes each
@@ -114,11 +122,11 @@ void py_dequantize_multi(at::Tensor comp, at::Tensor out, int64_t bits,
 }
 
 int64_t py_buffer_size(int64_t n, at::ScalarType st, int64_t bits,
-                       int64_t bucket_size) {
+                       int64_t bucket_size, bool skip_incomplete) {
   DType dt = st == at::kFloat ? DType::F32
              : st == at::kHalf ? DType::F16
                                : DType::BF16;
-  return buffer_size(n, dt, (int)bits, (int)bucket_size);
+  return buffer_size(n, dt, (int)bits, (int)bucket_size, skip_incomplete);
 }
 
 std::pair<std::vector<int64_t>, std::vector<int64_t>> py_partition(
@@ -167,12 +175,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
         py::arg("bucket_size"), py::arg("stochastic") = false,
-        py::arg("seed") = 0);
+        py::arg("seed") = 0, py::arg("skip_incomplete") = false);
   m.def("dequantize", &cgx::py_dequantize, py::arg("comp"), py::arg("out"),
-        py::arg("bits"), py::arg("bucket_size"), py::arg("add") = false);
+        py::arg("bits"), py::arg("bucket_size"), py::arg("add") = false,
+        py::arg("skip_incomplete") = false);
   m.def("dequantize_multi", &cgx::py_dequantize_multi, py::arg("comp"),
         py::arg("out"), py::arg("bits"), py::arg("bucket_size"),
         py::arg("add") = false);
-  m.def("buffer_size", &cgx::py_buffer_size);
+  m.def("buffer_size", &cgx::py_buffer_size, py::arg("n"), py::arg("dtype"),
+        py::arg("bits"), py::arg("bucket_size"),
+        py::arg("skip_incomplete") = false);
   m.def("partition", &cgx::py_partition);
 }

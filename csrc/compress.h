@@ -21,11 +21,23 @@ inline int elem_size(DType d) { return d == DType::F32 ? 4 : 2; }
 inline int64_t align8(int64_t x) { return (x + 7) / 8 * 8; }
 
 // Compressed byte size of an n-element slice (golden.buffer_size parity).
-inline int64_t buffer_size(int64_t n, DType dt, int bits, int bucket_size) {
+// skip_incomplete: the trailing partial bucket is stored as raw T values
+// after the aligned packed region instead of being quantized.
+inline int64_t buffer_size(int64_t n, DType dt, int bits, int bucket_size,
+                           bool skip_incomplete = false) {
   if (n == 0) return 0;
-  const int64_t nb = (n + bucket_size - 1) / bucket_size;
-  return 2 * nb * elem_size(dt) + align8((n * bits + 7) / 8);
+  int64_t nb = (n + bucket_size - 1) / bucket_size;
+  int64_t residuals = 0;
+  if (skip_incomplete) {
+    nb = n / bucket_size;
+    residuals = n % bucket_size;
+    n = nb * bucket_size;
+  }
+  return 2 * nb * elem_size(dt) + align8((n * bits + 7) / 8) +
+         residuals * elem_size(dt);
 }
+
+constexpr int32_t kFlagSkipIncomplete = 1;
 
 // One quantize work item: compress `n` elems at `in` into `out` bytes.
 struct QuantDesc {
@@ -33,7 +45,7 @@ struct QuantDesc {
   uint8_t* out;
   int64_t n;
   int32_t bucket;
-  int32_t pad_;
+  int32_t flags;  // kFlagSkipIncomplete
 };
 
 // One dequantize work item: decode `n` elems from `nsrc` compressed streams
@@ -47,7 +59,7 @@ struct DequantDesc {
   int32_t bucket;
   int32_t nsrc;
   int32_t add;
-  int32_t pad_;
+  int32_t flags;  // kFlagSkipIncomplete
 };
 
 // Batched launchers. descs/cum live in DEVICE memory. `cum` is the exclusive
@@ -57,14 +69,18 @@ struct DequantDesc {
 // All slices in one launch share `bits` (the engine groups by bits).
 // buckets_mult8: caller guarantees every slice's bucket_size % 8 == 0 (the
 // fused single-read path); otherwise a meta pass + generic pack pass run.
+// any_residual: at least one slice carries kFlagSkipIncomplete with a
+// non-empty raw residual tail (adds one small copy/accumulate pass).
 void launch_quantize_batch(const QuantDesc* descs, const int64_t* cum,
                            int nslices, int64_t total_buckets, DType dt,
                            int bits, uint64_t seed, bool stochastic,
-                           hipStream_t stream, bool buckets_mult8);
+                           hipStream_t stream, bool buckets_mult8,
+                           bool any_residual = false);
 
 void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
                              int nslices, int64_t total_groups, DType dt,
-                             int bits, hipStream_t stream);
+                             int bits, hipStream_t stream,
+                             bool any_residual = false);
 
 // y[i] += x[i] elementwise (T precision), n elements.
 void launch_add(const void* x, void* y, int64_t n, DType dt,

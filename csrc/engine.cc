@@ -109,6 +109,8 @@ EngineConfig EngineConfig::from_env() {
   c.default_bits = (int)env_int("CGX_COMPRESSION_QUANTIZATION_BITS", 32);
   c.default_bucket = (int)env_int("CGX_COMPRESSION_BUCKET_SIZE", 512);
   c.stochastic = env_int("CGX_STOCHASTIC_ROUNDING", 1) != 0;
+  c.skip_incomplete =
+      env_int("CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS", 0) != 0;
   const char* fr = std::getenv("CGX_COMPRESSION_FAKE_RATIO");
   if (fr && *fr) c.fake_ratio = std::atof(fr);
   if (!(c.fake_ratio > 0.0 && c.fake_ratio <= 1.0)) c.fake_ratio = 1.0;
@@ -313,16 +315,23 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
     auto* qd = reinterpret_cast<QuantDesc*>(host);
     auto* cum = reinterpret_cast<int64_t*>(host + desc_bytes);
     cum[0] = 0;
+    bool any_residual = false;
     for (int i = 0; i < nsl; i++) {
       const Slice& s = *list[i];
-      qd[i] = QuantDesc{s.data, out_base + s.comp_off, s.n, s.bucket, 0};
-      cum[i + 1] = cum[i] + (s.n + s.bucket - 1) / s.bucket;
+      const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
+      qd[i] = QuantDesc{s.data, out_base + s.comp_off, s.n, s.bucket, flags};
+      if (s.skip_incomplete) {
+        cum[i + 1] = cum[i] + s.n / s.bucket;
+        any_residual |= (s.n % s.bucket) != 0;
+      } else {
+        cum[i + 1] = cum[i] + (s.n + s.bucket - 1) / s.bucket;
+      }
     }
     char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
     launch_quantize_batch(reinterpret_cast<QuantDesc*>(dev),
                           reinterpret_cast<int64_t*>(dev + desc_bytes), nsl,
                           cum[nsl], dt, key.first, seed_++, stochastic, stream,
-                          key.second);
+                          key.second, any_residual);
   }
 }
 
@@ -342,21 +351,29 @@ void Engine::run_dequant(const std::vector<Slice>& slices,
     auto* dd = reinterpret_cast<DequantDesc*>(host);
     auto* cum = reinterpret_cast<int64_t*>(host + desc_bytes);
     cum[0] = 0;
+    bool any_residual = false;
     for (int i = 0; i < nsl; i++) {
       const Slice& s = *list[i];
+      const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
       dd[i] = DequantDesc{in_base + s.comp_off, s.data,    s.n, src_stride,
-                          s.bucket,             nsrc, add ? 1 : 0, 0};
-      cum[i + 1] = cum[i] + (s.n + 7) / 8;
+                          s.bucket,             nsrc, add ? 1 : 0, flags};
+      if (s.skip_incomplete) {
+        const int64_t nq = s.n / s.bucket * (int64_t)s.bucket;
+        cum[i + 1] = cum[i] + (nq + 7) / 8;
+        any_residual |= (s.n % s.bucket) != 0;
+      } else {
+        cum[i + 1] = cum[i] + (s.n + 7) / 8;
+      }
     }
     char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
     launch_dequantize_batch(reinterpret_cast<DequantDesc*>(dev),
                             reinterpret_cast<int64_t*>(dev + desc_bytes), nsl,
-                            cum[nsl], dt, bits, stream);
+                            cum[nsl], dt, bits, stream, any_residual);
   }
 }
 
 Engine::ChunkPlan Engine::plan(const std::vector<LayerView>& views,
-                               DType dt) {
+                               DType dt, bool skip_incomplete) {
   ChunkPlan pl;
   const int ws = size_;
   const int es = elem_size(dt);
@@ -378,8 +395,9 @@ Engine::ChunkPlan Engine::plan(const std::vector<LayerView>& views,
       const int64_t hi = std::min(pos + v.numel, end);
       if (hi > lo) {
         pl.rs[r].push_back(Slice{v.data + (lo - pos) * es, hi - lo, v.bits,
-                                 v.bucket_size, coff});
-        coff += buffer_size(hi - lo, dt, v.bits, v.bucket_size);
+                                 v.bucket_size, coff, skip_incomplete});
+        coff += buffer_size(hi - lo, dt, v.bits, v.bucket_size,
+                            skip_incomplete);
       }
       pos += v.numel;
       if (pos >= end) break;
@@ -393,7 +411,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
                        ncclComm_t comm, hipStream_t qs,
                        const EngineConfig& cfg) {
   const int ws = size_;
-  ChunkPlan pl = plan(views, dt);
+  ChunkPlan pl = plan(views, dt, cfg.skip_incomplete);
   if (pl.n == 0) return;
   auto& rs = pl.rs;
   auto& comp = pl.comp;
@@ -508,7 +526,7 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
   // the running partial sum, then ws-1 allgather steps FORWARDING the
   // once-quantized reduced segments, final batch decompress.
   const int ws = size_;
-  ChunkPlan pl = plan(views, dt);
+  ChunkPlan pl = plan(views, dt, cfg.skip_incomplete);
   if (pl.n == 0) return;
   auto& rs = pl.rs;
   auto& comp = pl.comp;

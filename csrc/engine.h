@@ -92,6 +92,7 @@ struct EngineConfig {
   bool stochastic = true;
   bool ring = false;  // CGX_INNER_REDUCTION_TYPE=Ring (default SRA)
   double fake_ratio = 1.0;  // CGX_COMPRESSION_FAKE_RATIO (bandwidth expt)
+  bool skip_incomplete = false;  // CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS
   static EngineConfig from_env();  // re-read every bucket like the reference
 };
 
@@ -153,6 +154,7 @@ class Engine {
     int bits;
     int bucket;
     int64_t comp_off;  // byte offset of this slice in the chunk's comp stream
+    bool skip_incomplete = false;
   };
 
   // Double-buffered staging so chunk c+1's quantize (on qs) can start while
@@ -169,7 +171,8 @@ class Engine {
     std::vector<int64_t> offs, szs;
     int64_t n = 0;
   };
-  ChunkPlan plan(const std::vector<LayerView>& views, DType dt);
+  ChunkPlan plan(const std::vector<LayerView>& views, DType dt,
+                 bool skip_incomplete);
 
   void sra_chunk(const std::vector<LayerView>& views, DType dt,
                  ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg);

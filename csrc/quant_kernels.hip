@@ -18,6 +18,7 @@
 #include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <cstdio>
 
 #include "compress.h"
@@ -214,10 +215,13 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
       if (cum[mid] <= b) lo = mid; else hi = mid;
     }
     const QuantDesc d = descs[lo];
+    const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
+    const int64_t nq =
+        skip ? (d.n / d.bucket) * static_cast<int64_t>(d.bucket) : d.n;
     const int64_t nb_slice = cum[lo + 1] - cum[lo];
     const int64_t lb = b - cum[lo];
     const int64_t bstart = lb * static_cast<int64_t>(d.bucket);
-    const int cur = static_cast<int>(min(static_cast<int64_t>(d.bucket), d.n - bstart));
+    const int cur = static_cast<int>(min(static_cast<int64_t>(d.bucket), nq - bstart));
     const T* in = reinterpret_cast<const T*>(d.in) + bstart;
     const bool al16 = (reinterpret_cast<uintptr_t>(in) & 15) == 0;
     const int ngroups = (cur + 7) >> 3;
@@ -261,7 +265,7 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     if constexpr (ENCODE) {
       uint8_t* packed =
           reinterpret_cast<uint8_t*>(d.out) + 2 * sizeof(R) * nb_slice;
-      const int64_t num_char = (d.n * BITS + 7) >> 3;
+      const int64_t num_char = (nq * BITS + 7) >> 3;
       const int64_t gbase = bstart >> 3;
       const bool live = unitf >= kEps;
       const float rinv = 1.0f / unitf;  // hoisted: fp32 div is 1/4 VALU rate
@@ -314,16 +318,20 @@ __global__ __launch_bounds__(kThreads) void k_pack_generic(
   constexpr float divisor = static_cast<float>((1 << BITS) - 1);
   for (int s = 0; s < nslices; s++) {
     const QuantDesc d = descs[s];
-    const int64_t ngroups = (d.n + 7) >> 3;
-    const int64_t nb_slice = (d.n + d.bucket - 1) / d.bucket;
+    const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
+    const int64_t nq =
+        skip ? (d.n / d.bucket) * static_cast<int64_t>(d.bucket) : d.n;
+    const int64_t ngroups = (nq + 7) >> 3;
+    const int64_t nb_slice =
+        skip ? d.n / d.bucket : (d.n + d.bucket - 1) / d.bucket;
     const R* meta = reinterpret_cast<const R*>(d.out);
     uint8_t* packed =
         reinterpret_cast<uint8_t*>(d.out) + 2 * sizeof(R) * nb_slice;
-    const int64_t num_char = (d.n * BITS + 7) >> 3;
+    const int64_t num_char = (nq * BITS + 7) >> 3;
     const R* in = reinterpret_cast<const R*>(d.in);
     const bool small = d.n < (int64_t(1) << 31);
     for (int64_t g = t0; g < ngroups; g += stride) {
-      const int m = static_cast<int>(min(static_cast<int64_t>(8), d.n - g * 8));
+      const int m = static_cast<int>(min(static_cast<int64_t>(8), nq - g * 8));
       uint64_t value = 0;
       const uint64_t pr =
           stochastic ? rand_pack(seed, (static_cast<uint64_t>(s) << 44) |
@@ -373,11 +381,15 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       if (cum[mid] <= w) lo = mid; else hi = mid;
     }
     const DequantDesc d = descs[lo];
+    const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
+    const int64_t nq =
+        skip ? (d.n / d.bucket) * static_cast<int64_t>(d.bucket) : d.n;
     const int64_t g = w - cum[lo];
-    const int64_t nb_slice = (d.n + d.bucket - 1) / d.bucket;
+    const int64_t nb_slice =
+        skip ? d.n / d.bucket : (d.n + d.bucket - 1) / d.bucket;
     const int64_t meta_bytes = 2 * sizeof(R) * nb_slice;
-    const int64_t num_char = (d.n * BITS + 7) >> 3;
-    const int m = static_cast<int>(min(static_cast<int64_t>(8), d.n - g * 8));
+    const int64_t num_char = (nq * BITS + 7) >> 3;
+    const int m = static_cast<int>(min(static_cast<int64_t>(8), nq - g * 8));
     const int64_t gb = g * BITS;
     const int nbytes =
         static_cast<int>(min(static_cast<int64_t>(BITS), num_char - gb));
@@ -438,6 +450,66 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
   }
 }
 
+// skip_incomplete residual handling: the trailing partial bucket travels as
+// raw T values after the aligned packed region (reference
+// MaxMinQuantizer::Compress/DecompressBuffer, compressor.cc:332-341,384-400).
+template <typename T>
+__global__ __launch_bounds__(kThreads) void k_residual_q(
+    const QuantDesc* __restrict__ descs, int nslices, int bits) {
+  using R = typename RawOf<T>::type;
+  const int64_t t0 =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int s = 0; s < nslices; s++) {
+    const QuantDesc d = descs[s];
+    if (!(d.flags & kFlagSkipIncomplete)) continue;
+    const int64_t nb = d.n / d.bucket;
+    const int64_t nq = nb * static_cast<int64_t>(d.bucket);
+    const int64_t r = d.n - nq;
+    if (r == 0) continue;
+    const R* src = reinterpret_cast<const R*>(d.in) + nq;
+    R* dst = reinterpret_cast<R*>(
+        d.out + 2 * sizeof(R) * nb + ((nq * bits + 7) / 8 + 7) / 8 * 8);
+    for (int64_t i = t0; i < r; i += stride) dst[i] = src[i];
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kThreads) void k_residual_d(
+    const DequantDesc* __restrict__ descs, int nslices, int bits) {
+  using R = typename RawOf<T>::type;
+  const int64_t t0 =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int s = 0; s < nslices; s++) {
+    const DequantDesc d = descs[s];
+    if (!(d.flags & kFlagSkipIncomplete)) continue;
+    const int64_t nb = d.n / d.bucket;
+    const int64_t nq = nb * static_cast<int64_t>(d.bucket);
+    const int64_t r = d.n - nq;
+    if (r == 0) continue;
+    const int64_t comp = 2 * sizeof(R) * nb + ((nq * bits + 7) / 8 + 7) / 8 * 8;
+    R* outp = reinterpret_cast<R*>(d.out) + nq;
+    for (int64_t i = t0; i < r; i += stride) {
+      uint32_t v;
+      bool have = d.add != 0;
+      if (have) v = outp[i];
+      for (int src = 0; src < d.nsrc; src++) {
+        const R* rp = reinterpret_cast<const R*>(d.in + src * d.src_stride +
+                                                 comp);
+        const uint32_t dec = rp[i];
+        if (!have && src == 0) {
+          v = dec;
+          have = true;
+        } else {
+          v = f2raw<T>(raw2f<T>(v) + raw2f<T>(dec));
+        }
+      }
+      outp[i] = static_cast<R>(v);
+    }
+  }
+}
+
 template <typename T>
 __global__ __launch_bounds__(kThreads) void k_add(const T* __restrict__ x,
                                                   T* __restrict__ y,
@@ -485,32 +557,48 @@ inline int grid_for(int64_t work, int per_block) {
 void launch_quantize_batch(const QuantDesc* descs, const int64_t* cum,
                            int nslices, int64_t total_buckets, DType dt,
                            int bits, uint64_t seed, bool stochastic,
-                           hipStream_t stream, bool buckets_mult8) {
-  if (total_buckets <= 0 || nslices <= 0) return;
-  const int grid = grid_for(total_buckets, kThreads / kWave);
+                           hipStream_t stream, bool buckets_mult8,
+                           bool any_residual) {
+  if (nslices <= 0) return;
+  const int grid = grid_for(std::max<int64_t>(total_buckets, 1),
+                            kThreads / kWave);
   CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
-    if (buckets_mult8) {
-      hipLaunchKernelGGL((k_quantize<T, BITS, true>), dim3(grid),
-                         dim3(kThreads), 0, stream, descs, cum, nslices,
-                         total_buckets, seed, (int)stochastic);
-    } else {
-      hipLaunchKernelGGL((k_quantize<T, BITS, false>), dim3(grid),
-                         dim3(kThreads), 0, stream, descs, cum, nslices,
-                         total_buckets, seed, (int)stochastic);
-      hipLaunchKernelGGL((k_pack_generic<T, BITS>), dim3(grid), dim3(kThreads),
-                         0, stream, descs, nslices, seed, (int)stochastic);
+    if (total_buckets > 0) {
+      if (buckets_mult8) {
+        hipLaunchKernelGGL((k_quantize<T, BITS, true>), dim3(grid),
+                           dim3(kThreads), 0, stream, descs, cum, nslices,
+                           total_buckets, seed, (int)stochastic);
+      } else {
+        hipLaunchKernelGGL((k_quantize<T, BITS, false>), dim3(grid),
+                           dim3(kThreads), 0, stream, descs, cum, nslices,
+                           total_buckets, seed, (int)stochastic);
+        hipLaunchKernelGGL((k_pack_generic<T, BITS>), dim3(grid),
+                           dim3(kThreads), 0, stream, descs, nslices, seed,
+                           (int)stochastic);
+      }
+    }
+    if (any_residual) {
+      hipLaunchKernelGGL((k_residual_q<T>), dim3(32), dim3(kThreads), 0,
+                         stream, descs, nslices, bits);
     }
   }));
 }
 
 void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
                              int nslices, int64_t total_groups, DType dt,
-                             int bits, hipStream_t stream) {
-  if (total_groups <= 0 || nslices <= 0) return;
-  const int grid = grid_for(total_groups, kThreads);
+                             int bits, hipStream_t stream,
+                             bool any_residual) {
+  if (nslices <= 0) return;
+  const int grid = grid_for(std::max<int64_t>(total_groups, 1), kThreads);
   CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
-    hipLaunchKernelGGL((k_dequant<T, BITS>), dim3(grid), dim3(kThreads), 0,
-                       stream, descs, cum, nslices, total_groups);
+    if (total_groups > 0) {
+      hipLaunchKernelGGL((k_dequant<T, BITS>), dim3(grid), dim3(kThreads), 0,
+                         stream, descs, cum, nslices, total_groups);
+    }
+    if (any_residual) {
+      hipLaunchKernelGGL((k_residual_d<T>), dim3(32), dim3(kThreads), 0,
+                         stream, descs, nslices, bits);
+    }
   }));
 }
 

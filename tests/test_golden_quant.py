@@ -82,3 +82,26 @@ def test_stochastic_rand_tensor():
     lv_sto = golden.encode_levels(x, meta, 4, 512, rand)
     # stochastic levels differ from deterministic by at most 1
     assert (np.abs(lv_det.astype(int) - lv_sto.astype(int)) <= 1).all()
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("n,bucket", [(1000, 512), (1025, 64), (7, 512),
+                                      (4096, 512)])
+def test_skip_incomplete_roundtrip(dtype, n, bucket):
+    torch.manual_seed(n)
+    x = torch.randn(n).to(dtype)
+    comp = golden.quantize(x, 4, bucket, skip_incomplete=True)
+    assert comp.numel() == golden.buffer_size(n, dtype, 4, bucket, True)
+    out = golden.dequantize(comp, n, dtype, 4, bucket, skip_incomplete=True)
+    r = n % bucket
+    if r:
+        # raw residual tail is exact
+        assert torch.equal(out[-r:], x[-r:])
+    nq = n - r
+    if nq:
+        xb = x[:nq].float().view(-1, bucket)
+        unit = (xb.max(1).values - xb.min(1).values) / 15
+        err = (xb - out[:nq].float().view(-1, bucket)).abs().max(1).values
+        tol = unit * 1.05 + 1e-6 + (0 if dtype == torch.float32 else
+                                    xb.abs().max() * 2 ** -7)
+        assert (err <= tol).all()

@@ -162,3 +162,36 @@ def test_dequantize_throughput_floor():
     gbps = n * 4 / dt / 1e9
     print(f"dequantize 64M fp32 4-bit: {dt*1e3:.2f} ms, {gbps:.0f} GB/s out")
     assert gbps > 1000, f"dequantize too slow: {gbps:.0f} GB/s"
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("bits", [2, 4, 8])
+@pytest.mark.parametrize("n,bucket", [(1000, 512), (1025, 64), (7, 512),
+                                      (4096, 512), (131, 8)])
+def test_skip_incomplete_matches_golden(dtype, bits, n, bucket):
+    from torch_cgx_amd import _C
+    torch.manual_seed(n + bits)
+    x = torch.randn(n).to(dtype)
+    comp_gold = golden.quantize(x, bits, bucket, skip_incomplete=True)
+    comp_gpu = _C.quantize(x.to(_dev()), bits, bucket, False, 0, True).cpu()
+    assert torch.equal(comp_gpu, comp_gold)
+    expected = golden.dequantize(comp_gold, n, dtype, bits, bucket,
+                                 skip_incomplete=True)
+    out = torch.empty(n, dtype=dtype, device=_dev())
+    _C.dequantize(comp_gold.to(_dev()), out, bits, bucket, False, True)
+    assert torch.equal(out.cpu(), expected)
+
+
+def test_skip_incomplete_dequant_add():
+    from torch_cgx_amd import _C
+    torch.manual_seed(9)
+    n, bits, bucket = 1025, 4, 64
+    x = torch.randn(n)
+    base = torch.randn(n)
+    comp = golden.quantize(x, bits, bucket, skip_incomplete=True)
+    dec = golden.dequantize(comp, n, torch.float32, bits, bucket,
+                            skip_incomplete=True)
+    expected = base + dec
+    out = base.to(_dev()).clone()
+    _C.dequantize(comp.to(_dev()), out, bits, bucket, True, True)
+    assert torch.equal(out.cpu(), expected)

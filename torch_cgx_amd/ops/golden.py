@@ -151,8 +151,22 @@ def unpack_levels(packed: np.ndarray, n: int, bits: int) -> np.ndarray:
     return levels[:n]
 
 
+def _raw_bytes(t: torch.Tensor) -> torch.Tensor:
+    raw = t.view(torch.int16) if t.dtype == torch.bfloat16 else t
+    return torch.from_numpy(
+        np.frombuffer(raw.contiguous().numpy().tobytes(),
+                      dtype=np.uint8).copy())
+
+
+def _from_raw_bytes(b: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    arr = b.numpy().copy()
+    if dtype == torch.float32:
+        return torch.from_numpy(arr.view(np.float32).copy())
+    return torch.from_numpy(arr.view(np.int16).copy()).view(dtype)
+
+
 def quantize(x: torch.Tensor, bits: int, bucket_size: int,
-             rand=0.5) -> torch.Tensor:
+             rand=0.5, skip_incomplete: bool = False) -> torch.Tensor:
     """Compress x -> uint8 buffer of exactly buffer_size(...) bytes.
 
     Alignment padding bytes are zero-filled (the reference leaves them
@@ -161,6 +175,19 @@ def quantize(x: torch.Tensor, bits: int, bucket_size: int,
     assert 1 <= bits <= 8
     x = x.contiguous().view(-1)
     n = x.numel()
+    if skip_incomplete:
+        nb = n // bucket_size
+        nq = nb * bucket_size
+        r = n - nq
+        total = buffer_size(n, x.dtype, bits, bucket_size, True)
+        out = torch.zeros(total, dtype=torch.uint8)
+        if nq:
+            head = quantize(x[:nq], bits, bucket_size, rand)
+            out[: head.numel()] = head
+        if r:
+            rb = _raw_bytes(x[nq:])
+            out[total - rb.numel():] = rb
+        return out
     total = buffer_size(n, x.dtype, bits, bucket_size)
     out = torch.zeros(total, dtype=torch.uint8)
     meta = compute_meta(x, bits, bucket_size)
@@ -176,9 +203,22 @@ def quantize(x: torch.Tensor, bits: int, bucket_size: int,
 
 
 def dequantize(buf: torch.Tensor, n: int, dtype: torch.dtype, bits: int,
-               bucket_size: int) -> torch.Tensor:
+               bucket_size: int, skip_incomplete: bool = False) -> torch.Tensor:
     """Decompress a quantize() buffer back to an n-element tensor of dtype."""
     buf = buf.contiguous().view(-1)
+    if skip_incomplete:
+        nb = n // bucket_size
+        nq = nb * bucket_size
+        r = n - nq
+        parts = []
+        if nq:
+            head_bytes = buffer_size(nq, dtype, bits, bucket_size)
+            parts.append(dequantize(buf[:head_bytes], nq, dtype, bits,
+                                    bucket_size))
+        if r:
+            es = elem_size(dtype)
+            parts.append(_from_raw_bytes(buf[buf.numel() - r * es:], dtype))
+        return torch.cat(parts) if parts else torch.zeros(0, dtype=dtype)
     nb = num_buckets(n, bucket_size)
     es = elem_size(dtype)
     meta_bytes = 2 * nb * es
