@@ -45,6 +45,42 @@ __device__ __forceinline__ float rand_lane(uint64_t pack_rand, int j) {
   return static_cast<float>((pack_rand >> (8 * j)) & 0xFF) * (1.0f / 256.0f);
 }
 
+// Full-wave min/max reduction via DPP row operations (register path, ~2
+// cycles/step) instead of __shfl_xor, which hipcc lowers to six dependent
+// ds_bpermute_b32 (LDS latency) per value -- the dominant cost of the
+// original wave-per-bucket kernel.
+// Sequence: row_shr 1/2/4/8 then row_bcast15/31; lane 63 holds the result.
+#define CGX_DPP_STEP(CTRL, OP)                                                t = __builtin_amdgcn_update_dpp(iid, __builtin_bit_cast(int, v), (CTRL),                                    0xf, 0xf, false);                           v = OP(v, __builtin_bit_cast(float, t));
+
+__device__ __forceinline__ void wave_minmax(float& vmin, float& vmax) {
+  int t;
+  {
+    const int iid = __builtin_bit_cast(int, INFINITY);
+    float v = vmin;
+    CGX_DPP_STEP(0x111, fminf)  // row_shr:1
+    CGX_DPP_STEP(0x112, fminf)  // row_shr:2
+    CGX_DPP_STEP(0x114, fminf)  // row_shr:4
+    CGX_DPP_STEP(0x118, fminf)  // row_shr:8
+    CGX_DPP_STEP(0x142, fminf)  // row_bcast:15
+    CGX_DPP_STEP(0x143, fminf)  // row_bcast:31
+    vmin = __builtin_bit_cast(
+        float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, v), 63));
+  }
+  {
+    const int iid = __builtin_bit_cast(int, -INFINITY);
+    float v = vmax;
+    CGX_DPP_STEP(0x111, fmaxf)
+    CGX_DPP_STEP(0x112, fmaxf)
+    CGX_DPP_STEP(0x114, fmaxf)
+    CGX_DPP_STEP(0x118, fmaxf)
+    CGX_DPP_STEP(0x142, fmaxf)
+    CGX_DPP_STEP(0x143, fmaxf)
+    vmax = __builtin_bit_cast(
+        float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, v), 63));
+  }
+}
+#undef CGX_DPP_STEP
+
 template <typename To, typename From>
 __device__ __forceinline__ To bitcast(From f) {
   static_assert(sizeof(To) == sizeof(From));
@@ -228,31 +264,41 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
 
     uint32_t stash[MAXG][8];
     float lmin = INFINITY, lmax = -INFINITY;
-    for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
-      uint32_t r[8];
-      const int m = min(8, cur - g * 8);
-      if (m == 8) {
-        load8<T>(in + g * 8, al16, r);
-      } else {
-        const R* q = reinterpret_cast<const R*>(in) + g * 8;
-        for (int j = 0; j < m; j++) r[j] = q[j];
-        for (int j = m; j < 8; j++) r[j] = 0;
-      }
-      if (gi < MAXG) {
+    // fast path: full bucket, whole groups per lane (wave-uniform branch)
+    const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
+    if (full && ngroups <= MAXG * kWave) {
+      for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+        load8<T>(in + g * 8, true, stash[gi]);
 #pragma unroll
-        for (int j = 0; j < 8; j++) stash[gi][j] = r[j];
+        for (int j = 0; j < 8; j++) {
+          const float f = raw2f<T>(stash[gi][j]);
+          lmin = fminf(lmin, f);
+          lmax = fmaxf(lmax, f);
+        }
       }
-      for (int j = 0; j < m; j++) {
-        const float f = raw2f<T>(r[j]);
-        lmin = fminf(lmin, f);
-        lmax = fmaxf(lmax, f);
+    } else {
+      for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+        uint32_t r[8];
+        const int m = min(8, cur - g * 8);
+        if (m == 8) {
+          load8<T>(in + g * 8, al16, r);
+        } else {
+          const R* q = reinterpret_cast<const R*>(in) + g * 8;
+          for (int j = 0; j < m; j++) r[j] = q[j];
+          for (int j = m; j < 8; j++) r[j] = 0;
+        }
+        if (gi < MAXG) {
+#pragma unroll
+          for (int j = 0; j < 8; j++) stash[gi][j] = r[j];
+        }
+        for (int j = 0; j < m; j++) {
+          const float f = raw2f<T>(r[j]);
+          lmin = fminf(lmin, f);
+          lmax = fmaxf(lmax, f);
+        }
       }
     }
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      lmin = fminf(lmin, __shfl_xor(lmin, off));
-      lmax = fmaxf(lmax, __shfl_xor(lmax, off));
-    }
+    wave_minmax(lmin, lmax);
     const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
     const float unitf = raw2f<T>(unit_raw);
     const float minf = lmin;
@@ -269,6 +315,27 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
       const int64_t gbase = bstart >> 3;
       const bool live = unitf >= kEps;
       const float rinv = 1.0f / unitf;  // hoisted: fp32 div is 1/4 VALU rate
+      if (full && ngroups <= MAXG * kWave && live) {
+        for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+          const uint64_t pr =
+              stochastic
+                  ? rand_pack(seed, (static_cast<uint64_t>(lo) << 44) |
+                                        static_cast<uint64_t>(gbase + g))
+                  : 0;
+          uint64_t value = 0;
+#pragma unroll
+          for (int j = 0; j < 8; j++) {
+            const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
+            const float dd = (raw2f<T>(stash[gi][j]) - minf) * rinv + rnd;
+            const uint32_t level =
+                static_cast<uint32_t>(fminf(floorf(dd), divisor));
+            value |= static_cast<uint64_t>(level & ((1u << BITS) - 1))
+                     << (j * BITS);
+          }
+          store_bytes(packed + (gbase + g) * BITS, value, BITS);
+        }
+        continue;
+      }
       for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
         uint32_t r[8];
         const int m = min(8, cur - g * 8);
