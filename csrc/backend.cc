@@ -5,6 +5,7 @@
 #include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
 
 #include <cstring>
+#include <unistd.h>
 
 namespace cgx {
 
@@ -81,6 +82,44 @@ c10::intrusive_ptr<c10::ivalue::Future> WorkCGX::getFuture() {
 }
 
 // ---------------------------------------------------------------------------
+// Topology
+// ---------------------------------------------------------------------------
+Topology compute_topology(const c10::intrusive_ptr<c10d::Store>& store,
+                          int rank, int size, const std::string& hostname) {
+  store->set("cgx/host/" + std::to_string(rank),
+             std::vector<uint8_t>(hostname.begin(), hostname.end()));
+  std::vector<std::string> hosts(size);
+  for (int r = 0; r < size; r++) {
+    auto v = store->get("cgx/host/" + std::to_string(r));  // blocks until set
+    hosts[r] = std::string(v.begin(), v.end());
+  }
+  Topology t;
+  std::vector<std::string> node_order;  // node hostnames by first appearance
+  std::vector<int> node_of(size);
+  for (int r = 0; r < size; r++) {
+    int id = -1;
+    for (size_t i = 0; i < node_order.size(); i++)
+      if (node_order[i] == hosts[r]) { id = (int)i; break; }
+    if (id < 0) {
+      id = (int)node_order.size();
+      node_order.push_back(hosts[r]);
+    }
+    node_of[r] = id;
+  }
+  t.n_nodes = (int)node_order.size();
+  t.node_id = node_of[rank];
+  std::vector<int> counts(t.n_nodes, 0);
+  for (int r = 0; r < size; r++) {
+    if (r == rank) t.local_rank = counts[node_of[r]];
+    counts[node_of[r]]++;
+  }
+  t.local_size = counts[t.node_id];
+  for (int c : counts)
+    if (c != counts[0]) t.uniform = false;
+  return t;
+}
+
+// ---------------------------------------------------------------------------
 // ProcessGroupCGX
 // ---------------------------------------------------------------------------
 ProcessGroupCGX::ProcessGroupCGX(c10::intrusive_ptr<c10d::Store> store,
@@ -90,6 +129,8 @@ ProcessGroupCGX::ProcessGroupCGX(c10::intrusive_ptr<c10d::Store> store,
       cpu_(std::move(cpu_delegate)) {}
 
 ProcessGroupCGX::~ProcessGroupCGX() {
+  if (intra_comm_) (void)ncclCommDestroy(intra_comm_);
+  if (cross_comm_) (void)ncclCommDestroy(cross_comm_);
   if (comm_) (void)ncclCommDestroy(comm_);
   if (start_ev_) (void)hipEventDestroy(start_ev_);
 }
@@ -121,6 +162,51 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
       /*isHighPriority=*/true, device_index_);
   engine_ = std::make_unique<Engine>(rank_, size_);
   CGX_HIP_CHECK(hipEventCreateWithFlags(&start_ev_, hipEventDisableTiming));
+
+  // multi-node: build intra/cross communicators for the hierarchical path
+  // (CGX_HIERARCHICAL=0 disables; CGX_INTRA_BROADCAST=0 falls back flat,
+  // matching the reference's mode switch)
+  char host[256] = {0};
+  gethostname(host, sizeof(host) - 1);
+  topo_ = compute_topology(store_, rank_, size_, host);
+  const char* henv = std::getenv("CGX_HIERARCHICAL");
+  const bool want_hier = !(henv && std::strcmp(henv, "0") == 0);
+  if (topo_.n_nodes > 1 && topo_.uniform && topo_.local_size > 1 &&
+      want_hier) {
+    // intra communicator: ranks on this node
+    const std::string ikey = "cgx/intra_uid/" + std::to_string(topo_.node_id);
+    ncclUniqueId iid;
+    if (topo_.local_rank == 0) {
+      CGX_NCCL_CHECK(ncclGetUniqueId(&iid));
+      store_->set(ikey, std::vector<uint8_t>(
+                            reinterpret_cast<uint8_t*>(&iid),
+                            reinterpret_cast<uint8_t*>(&iid) + sizeof(iid)));
+    } else {
+      auto v = store_->get(ikey);
+      std::memcpy(&iid, v.data(), sizeof(iid));
+    }
+    CGX_NCCL_CHECK(ncclCommInitRank(&intra_comm_, topo_.local_size, iid,
+                                    topo_.local_rank));
+    // cross communicator: peers with the same local_rank on every node
+    const std::string ckey =
+        "cgx/cross_uid/" + std::to_string(topo_.local_rank);
+    ncclUniqueId cid;
+    if (topo_.node_id == 0) {
+      CGX_NCCL_CHECK(ncclGetUniqueId(&cid));
+      store_->set(ckey, std::vector<uint8_t>(
+                            reinterpret_cast<uint8_t*>(&cid),
+                            reinterpret_cast<uint8_t*>(&cid) + sizeof(cid)));
+    } else {
+      auto v = store_->get(ckey);
+      std::memcpy(&cid, v.data(), sizeof(cid));
+    }
+    CGX_NCCL_CHECK(
+        ncclCommInitRank(&cross_comm_, topo_.n_nodes, cid, topo_.node_id));
+    intra_engine_ = std::make_unique<Engine>(topo_.local_rank,
+                                             topo_.local_size);
+    cross_engine_ = std::make_unique<Engine>(topo_.node_id, topo_.n_nodes);
+    hierarchical_ = true;
+  }
 }
 
 template <typename Fn>
@@ -172,7 +258,22 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
     CGX_HIP_CHECK(hipEventRecord(start_ev_, cur.stream()));
     CGX_HIP_CHECK(hipStreamWaitEvent(stream_->stream(), start_ev_, 0));
     hipStream_t fin = stream_->stream();
-    if (size_ > 1) fin = engine_->allreduce(t, comm_, stream_->stream());
+    if (size_ > 1) {
+      const char* ib = std::getenv("CGX_INTRA_BROADCAST");
+      const bool hier = hierarchical_ && !(ib && std::strcmp(ib, "0") == 0);
+      if (hier) {
+        // intra-node compressed allreduce over xGMI, cross-node reduction on
+        // node leaders, then intra broadcast of the result
+        fin = intra_engine_->allreduce(t, intra_comm_, stream_->stream());
+        if (topo_.local_rank == 0)
+          fin = cross_engine_->allreduce(t, cross_comm_, fin);
+        CGX_NCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                                     nccl_dtype(t), /*root=*/0, intra_comm_,
+                                     fin));
+      } else {
+        fin = engine_->allreduce(t, comm_, stream_->stream());
+      }
+    }
     auto fin_masq = c10::hip::getStreamFromExternalMasqueradingAsCUDA(
         fin, device_index_);
     c10::hip::HIPStreamGuardMasqueradingAsCUDA sguard(fin_masq.unwrap());

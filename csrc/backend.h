@@ -54,6 +54,20 @@ class WorkCGX : public c10d::Work {
   c10::intrusive_ptr<c10::ivalue::Future> future_;
 };
 
+// Rank topology computed by exchanging hostnames through the c10d Store
+// (replaces the reference's MPI_Comm_split_type node detection,
+// mpi_context.cc:25-35).
+struct Topology {
+  int node_id = 0;      // index of this rank's node (order of first rank)
+  int local_rank = 0;   // rank within the node
+  int local_size = 1;   // ranks on this node
+  int n_nodes = 1;
+  bool uniform = true;  // every node has the same local_size
+};
+
+Topology compute_topology(const c10::intrusive_ptr<c10d::Store>& store,
+                          int rank, int size, const std::string& hostname);
+
 class ProcessGroupCGX : public c10d::Backend {
  public:
   ProcessGroupCGX(c10::intrusive_ptr<c10d::Store> store, int rank, int size,
@@ -125,6 +139,15 @@ class ProcessGroupCGX : public c10d::Backend {
   ncclComm_t comm_ = nullptr;
   int device_index_ = -1;
   std::unique_ptr<Engine> engine_;
+  // hierarchical (multi-node) mode: intra-node SRA + cross-node reduction on
+  // node leaders + intra broadcast (reference CGX_INTRA_BROADCAST semantics,
+  // mpi_allreduce_operations.cc:160-183); single-node -> flat path.
+  Topology topo_;
+  ncclComm_t intra_comm_ = nullptr;
+  ncclComm_t cross_comm_ = nullptr;
+  std::unique_ptr<Engine> intra_engine_;
+  std::unique_ptr<Engine> cross_engine_;
+  bool hierarchical_ = false;
   std::optional<c10::hip::HIPStreamMasqueradingAsCUDA> stream_;
   hipEvent_t start_ev_ = nullptr;
   std::mutex mu_;

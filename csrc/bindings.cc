@@ -182,6 +182,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_multi", &cgx::py_dequantize_multi, py::arg("comp"),
         py::arg("out"), py::arg("bits"), py::arg("bucket_size"),
         py::arg("add") = false);
+  m.def("compute_topology",
+        [](const c10::intrusive_ptr<c10d::Store>& store, int rank, int size,
+           const std::string& hostname) {
+          auto t = cgx::compute_topology(store, rank, size, hostname);
+          return std::make_tuple(t.node_id, t.local_rank, t.local_size,
+                                 t.n_nodes, t.uniform);
+        },
+        "Exchange hostnames through the store -> (node_id, local_rank, "
+        "local_size, n_nodes, uniform)");
   m.def("buffer_size", &cgx::py_buffer_size, py::arg("n"), py::arg("dtype"),
         py::arg("bits"), py::arg("bucket_size"),
         py::arg("skip_incomplete") = false);
