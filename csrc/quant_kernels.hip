@@ -220,8 +220,23 @@ __device__ __forceinline__ void store_bytes(uint8_t* p, uint64_t v, int nb) {
 
 __device__ __forceinline__ uint64_t load_bytes(const uint8_t* p, int nb) {
   const uintptr_t a = reinterpret_cast<uintptr_t>(p);
-  if (nb == 8 && (a & 7) == 0) return *reinterpret_cast<const uint64_t*>(p);
-  if (nb == 4 && (a & 3) == 0) return *reinterpret_cast<const uint32_t*>(p);
+  if (nb == 8) {
+    if ((a & 7) == 0) return *reinterpret_cast<const uint64_t*>(p);
+    if ((a & 3) == 0) {
+      const uint32_t lo = reinterpret_cast<const uint32_t*>(p)[0];
+      const uint32_t hi = reinterpret_cast<const uint32_t*>(p)[1];
+      return static_cast<uint64_t>(lo) | (static_cast<uint64_t>(hi) << 32);
+    }
+  }
+  if (nb == 4) {
+    if ((a & 3) == 0) return *reinterpret_cast<const uint32_t*>(p);
+    if ((a & 1) == 0) {
+      const uint16_t lo = reinterpret_cast<const uint16_t*>(p)[0];
+      const uint16_t hi = reinterpret_cast<const uint16_t*>(p)[1];
+      return static_cast<uint64_t>(lo) | (static_cast<uint64_t>(hi) << 16);
+    }
+  }
+  if (nb == 2 && (a & 1) == 0) return *reinterpret_cast<const uint16_t*>(p);
   uint64_t v = 0;
   for (int i = 0; i < nb; i++) v |= static_cast<uint64_t>(p[i]) << (8 * i);
   return v;
@@ -437,82 +452,92 @@ template <typename T, int BITS>
 __global__ __launch_bounds__(kThreads) void k_dequant(
     const DequantDesc* __restrict__ descs, const int64_t* __restrict__ cum,
     int nslices, int64_t total_groups) {
+  // Outer loop over slices with every per-slice quantity hoisted; the inner
+  // grid-stride loop over FULL groups is branch-free (no tail/alignment
+  // checks, no exec-mask divergence) -- the original per-group binary-search
+  // form spent >80% of its instructions on repeated int64 address math.
   using R = typename RawOf<T>::type;
+  (void)cum;
+  (void)total_groups;
   const int64_t t0 =
       static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
   const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
-  for (int64_t w = t0; w < total_groups; w += stride) {
-    int lo = 0, hi = nslices;
-    while (hi - lo > 1) {
-      const int mid = (lo + hi) >> 1;
-      if (cum[mid] <= w) lo = mid; else hi = mid;
-    }
-    const DequantDesc d = descs[lo];
+  for (int si = 0; si < nslices; si++) {
+    const DequantDesc d = descs[si];
     const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
     const int64_t nq =
         skip ? (d.n / d.bucket) * static_cast<int64_t>(d.bucket) : d.n;
-    const int64_t g = w - cum[lo];
     const int64_t nb_slice =
         skip ? d.n / d.bucket : (d.n + d.bucket - 1) / d.bucket;
     const int64_t meta_bytes = 2 * sizeof(R) * nb_slice;
-    const int64_t num_char = (nq * BITS + 7) >> 3;
-    const int m = static_cast<int>(min(static_cast<int64_t>(8), nq - g * 8));
-    const int64_t gb = g * BITS;
-    const int nbytes =
-        static_cast<int>(min(static_cast<int64_t>(BITS), num_char - gb));
-    T* outp = reinterpret_cast<T*>(d.out) + g * 8;
-    const bool al16 =
-        m == 8 && (reinterpret_cast<uintptr_t>(outp) & 15) == 0;
-    const bool one_bucket = (d.bucket & 7) == 0;
-    const bool small = d.n < (int64_t(1) << 31);
+    const int64_t full_groups = nq >> 3;
+    const bool oneb = (d.bucket & 7) == 0;
+    const uint32_t B8 = oneb ? static_cast<uint32_t>(d.bucket >> 3) : 1u;
+    const bool al16 = (reinterpret_cast<uintptr_t>(d.out) & 15) == 0;
+    const bool small = nq < (int64_t(1) << 28);
+    T* const out_base = reinterpret_cast<T*>(d.out);
+    const uint8_t* const in0 = d.in + meta_bytes;  // packed base, source 0
 
-    uint32_t v[8];
-    bool have = d.add != 0;
-    if (have) {
-      if (al16) {
-        load8<T>(outp, true, v);
-      } else {
-        const R* q = reinterpret_cast<const R*>(outp);
-        for (int j = 0; j < m; j++) v[j] = q[j];
-      }
-    }
-    for (int s = 0; s < d.nsrc; s++) {
-      const uint8_t* src = d.in + s * d.src_stride;
-      const R* meta = reinterpret_cast<const R*>(src);
-      const uint64_t value = load_bytes(src + meta_bytes + gb, nbytes);
-      const int64_t bk0 =
-          one_bucket
-              ? (small ? static_cast<int64_t>(
-                             static_cast<uint32_t>(g * 8) /
-                             static_cast<uint32_t>(d.bucket))
-                       : (g * 8) / d.bucket)
-              : 0;
-      for (int j = 0; j < m; j++) {
-        const int64_t bk =
-            one_bucket ? bk0
-                       : (small ? static_cast<int64_t>(
-                                      static_cast<uint32_t>(g * 8 + j) /
-                                      static_cast<uint32_t>(d.bucket))
-                                : (g * 8 + j) / d.bucket);
-        const uint32_t lvl =
-            static_cast<uint32_t>((value >> (j * BITS)) & ((1u << BITS) - 1));
-        const float unitf = raw2f<T>(meta[2 * bk]);
-        const float minf = raw2f<T>(meta[2 * bk + 1]);
-        // decode in T precision: round(unit*lvl) then round(min + product)
-        const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
-        const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
-        if (!have && s == 0) {
-          v[j] = dec;
-        } else {
-          v[j] = f2raw<T>(raw2f<T>(v[j]) + raw2f<T>(dec));
+    for (int64_t g = t0; g < full_groups; g += stride) {
+      uint32_t v[8];
+      bool have = d.add != 0;
+      T* outp = out_base + g * 8;
+      if (have) load8<T>(outp, al16, v);
+      const uint32_t bk0 =
+          oneb ? (small ? static_cast<uint32_t>(g) / B8
+                        : static_cast<uint32_t>(g / static_cast<int64_t>(B8)))
+               : 0;
+      for (int sidx = 0; sidx < d.nsrc; sidx++) {
+        const uint8_t* src = in0 + sidx * d.src_stride;
+        const R* meta = reinterpret_cast<const R*>(src - meta_bytes);
+        const uint64_t value = load_bytes(src + g * BITS, BITS);
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          const int64_t bk = oneb ? bk0 : (g * 8 + j) / d.bucket;
+          const uint32_t lvl = static_cast<uint32_t>((value >> (j * BITS)) &
+                                                     ((1u << BITS) - 1));
+          const float unitf = raw2f<T>(meta[2 * bk]);
+          const float minf = raw2f<T>(meta[2 * bk + 1]);
+          const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
+          const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+          if (!have && sidx == 0) {
+            v[j] = dec;
+          } else {
+            v[j] = f2raw<T>(raw2f<T>(v[j]) + raw2f<T>(dec));
+          }
         }
       }
+      store8<T>(outp, al16, v);
     }
-    if (al16) {
-      store8<T>(outp, true, v);
-    } else {
-      R* q = reinterpret_cast<R*>(outp);
-      for (int j = 0; j < m; j++) q[j] = static_cast<R>(v[j]);
+
+    // tail group (fewer than 8 elements): one thread, scalar
+    const int mtail = static_cast<int>(nq - full_groups * 8);
+    if (mtail > 0 && t0 == 0) {
+      const int64_t g = full_groups;
+      const int64_t num_char = (nq * BITS + 7) >> 3;
+      const int nbytes = static_cast<int>(
+          min(static_cast<int64_t>(BITS), num_char - g * BITS));
+      R* outp = reinterpret_cast<R*>(d.out) + g * 8;
+      for (int sidx = 0; sidx < d.nsrc; sidx++) {
+        const uint8_t* src = in0 + sidx * d.src_stride;
+        const R* meta = reinterpret_cast<const R*>(src - meta_bytes);
+        const uint64_t value = load_bytes(src + g * BITS, nbytes);
+        for (int j = 0; j < mtail; j++) {
+          const int64_t bk = (g * 8 + j) / d.bucket;
+          const uint32_t lvl = static_cast<uint32_t>((value >> (j * BITS)) &
+                                                     ((1u << BITS) - 1));
+          const float unitf = raw2f<T>(meta[2 * bk]);
+          const float minf = raw2f<T>(meta[2 * bk + 1]);
+          const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
+          const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+          if (sidx == 0 && !d.add) {
+            outp[j] = static_cast<R>(dec);
+          } else {
+            outp[j] = static_cast<R>(
+                f2raw<T>(raw2f<T>(outp[j]) + raw2f<T>(dec)));
+          }
+        }
+      }
     }
   }
 }
