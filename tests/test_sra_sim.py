@@ -71,3 +71,32 @@ def test_tiny_chunks_more_ranks_than_elems():
     expected = torch.full((n,), float(sum(range(1, ws + 1))))
     for o in out:
         assert torch.allclose(o, expected)
+
+
+@pytest.mark.parametrize("ws", [3, 4])
+def test_ring_equal_inputs_exact(ws):
+    for n in [128, 4096]:
+        tensors = [torch.full((n,), 2.5) for _ in range(ws)]
+        out = sra_sim.ring_allreduce(tensors, [n], [(4, 512)])
+        expected = torch.full((n,), 2.5 * ws)
+        for o in out:
+            assert torch.equal(o, expected)
+
+
+@pytest.mark.parametrize("ws", [3, 4])
+@pytest.mark.parametrize("bits", [4, 8])
+def test_ring_error_bound(ws, bits):
+    n, bucket = 16384, 512
+    arange = np.arange(-n / 2, n / 2, 1.0)
+    tensors = [torch.tensor((r + 1) * arange, dtype=torch.float32)
+               for r in range(ws)]
+    expected = torch.tensor((ws * (ws + 1) / 2) * arange, dtype=torch.float32)
+    out = sra_sim.ring_allreduce(tensors, [n], [(bits, bucket)])
+    # per-hop requantize: looser bound than SRA (ws-1 quantizations of
+    # growing partial sums + the final one)
+    bound = 2 * bucket / ((1 << bits) - 1) * ws * (ws + 1) * ws
+    for o in out:
+        err = (o - expected).abs().max().item()
+        assert err < bound, (err, bound)
+    for o in out[1:]:
+        assert torch.equal(o, out[0])

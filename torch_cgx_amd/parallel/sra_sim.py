@@ -91,3 +91,43 @@ def sra_allreduce(tensors: List[torch.Tensor],
                              offsets[k], sizes[k], add=False)
 
     return [o.view(tensors[i].shape) for i, o in enumerate(out)]
+
+
+def ring_allreduce(tensors: List[torch.Tensor],
+                   layer_numels: Sequence[int],
+                   layer_configs: Sequence[Tuple[int, int]],
+                   rand=0.5) -> List[torch.Tensor]:
+    """Simulate the compressed Ring allreduce (csrc engine ring_chunk parity;
+    reference ring.cc:139-226): ws-1 reduce-scatter hops with per-hop
+    requantize of the running partial sum, then allgather hops forwarding the
+    once-quantized reduced segments."""
+    ws = len(tensors)
+    dtype = tensors[0].dtype
+    n = tensors[0].numel()
+    offsets, sizes = P.partition(n, ws, 0, layer_numels, dtype)
+    out = [t.clone().view(-1) for t in tensors]
+
+    # reduce-scatter: rank r sends chunk (r-s)%ws to r+1, accumulates
+    # chunk (r-s-1)%ws received from r-1
+    for s in range(ws - 1):
+        comps = {}
+        for r in range(ws):
+            sc = (r - s) % ws
+            comps[r] = compress_chunk(out[r], layer_numels, layer_configs,
+                                      offsets[sc], sizes[sc], rand)
+        for r in range(ws):
+            rc = (r - s - 1) % ws
+            decompress_chunk(comps[(r - 1) % ws], out[r], layer_numels,
+                             layer_configs, offsets[rc], sizes[rc], add=True)
+
+    # allgather: chunk k's fully-reduced segment lives on rank (k-1)%ws;
+    # its quantized bytes are forwarded unchanged and decoded everywhere
+    for k in range(ws):
+        producer = (k - 1) % ws
+        comp = compress_chunk(out[producer], layer_numels, layer_configs,
+                              offsets[k], sizes[k], rand)
+        for r in range(ws):
+            decompress_chunk(comp, out[r], layer_numels, layer_configs,
+                             offsets[k], sizes[k], add=False)
+
+    return [o.view(tensors[i].shape) for i, o in enumerate(out)]
