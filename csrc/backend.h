@@ -33,8 +33,7 @@ namespace cgx {
 class WorkCGX : public c10d::Work {
  public:
   WorkCGX(int rank, c10d::OpType op, at::Device device,
-          std::vector<at::Tensor> outputs,
-          const char* profiling_title = "cgx");
+          std::vector<at::Tensor> outputs);
   ~WorkCGX() override;
 
   // record the completion event on `stream` and mark the future completed

@@ -478,74 +478,9 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
     T* const out_base = reinterpret_cast<T*>(d.out);
     const uint8_t* const in0 = d.in + meta_bytes;  // packed base, source 0
 
-    // main loop: PAIRS of full groups per thread per iteration -- two
-    // independent decode chains (latency hiding at max occupancy), 2x wider
-    // packed loads and a 64B contiguous store span per thread
-    const int64_t npairs = full_groups >> 1;
-    for (int64_t p2 = t0; p2 < npairs; p2 += stride) {
-      const int64_t g = p2 * 2;
-      uint32_t v0[8], v1[8];
-      const bool have = d.add != 0;
-      T* outp = out_base + g * 8;
-      if (have) {
-        load8<T>(outp, al16, v0);
-        load8<T>(outp + 8, al16, v1);
-      }
-      uint32_t bk0 = 0, bk1 = 0;
-      if (oneb) {
-        if (small) {
-          const uint32_t gg = static_cast<uint32_t>(g);
-          bk0 = gg / B8;
-          bk1 = bk0 + ((gg + 1) - bk0 * B8 == B8 ? 1u : 0u);
-        } else {
-          bk0 = static_cast<uint32_t>(g / static_cast<int64_t>(B8));
-          bk1 = static_cast<uint32_t>((g + 1) / static_cast<int64_t>(B8));
-        }
-      }
-      for (int sidx = 0; sidx < d.nsrc; sidx++) {
-        const uint8_t* src = in0 + sidx * d.src_stride;
-        const R* meta = reinterpret_cast<const R*>(src - meta_bytes);
-        const uint64_t val01 = load_bytes(src + g * BITS, 2 * BITS <= 8
-                                                              ? 2 * BITS
-                                                              : BITS);
-        uint64_t value0 = val01, value1;
-        if (2 * BITS <= 8) {
-          value1 = val01 >> (8 * BITS);
-        } else {
-          value1 = load_bytes(src + (g + 1) * BITS, BITS);
-        }
-#pragma unroll
-        for (int j = 0; j < 8; j++) {
-          const int64_t bka = oneb ? bk0 : (g * 8 + j) / d.bucket;
-          const int64_t bkb = oneb ? bk1 : (g * 8 + 8 + j) / d.bucket;
-          const uint32_t la = static_cast<uint32_t>((value0 >> (j * BITS)) &
-                                                    ((1u << BITS) - 1));
-          const uint32_t lb = static_cast<uint32_t>((value1 >> (j * BITS)) &
-                                                    ((1u << BITS) - 1));
-          const uint32_t pa =
-              f2raw<T>(raw2f<T>(meta[2 * bka]) * static_cast<float>(la));
-          const uint32_t pb =
-              f2raw<T>(raw2f<T>(meta[2 * bkb]) * static_cast<float>(lb));
-          const uint32_t da =
-              f2raw<T>(raw2f<T>(meta[2 * bka + 1]) + raw2f<T>(pa));
-          const uint32_t db =
-              f2raw<T>(raw2f<T>(meta[2 * bkb + 1]) + raw2f<T>(pb));
-          if (!have && sidx == 0) {
-            v0[j] = da;
-            v1[j] = db;
-          } else {
-            v0[j] = f2raw<T>(raw2f<T>(v0[j]) + raw2f<T>(da));
-            v1[j] = f2raw<T>(raw2f<T>(v1[j]) + raw2f<T>(db));
-          }
-        }
-      }
-      store8<T>(outp, al16, v0);
-      store8<T>(outp + 8, al16, v1);
-    }
-    // leftover single full group (odd count)
-    for (int64_t g = npairs * 2 + t0; g < full_groups; g += stride) {
+    for (int64_t g = t0; g < full_groups; g += stride) {
       uint32_t v[8];
-      const bool have = d.add != 0;
+      bool have = d.add != 0;
       T* outp = out_base + g * 8;
       if (have) load8<T>(outp, al16, v);
       const uint32_t bk0 =
