@@ -39,8 +39,9 @@ void check_single(const std::vector<at::Tensor>& ts) {
 // WorkCGX
 // ---------------------------------------------------------------------------
 WorkCGX::WorkCGX(int rank, c10d::OpType op, at::Device device,
-                 std::vector<at::Tensor> outputs)
-    : c10d::Work(rank, op, "cgx"), device_(device),
+                 std::vector<at::Tensor> outputs,
+                 const char* profiling_title)
+    : c10d::Work(rank, op, profiling_title), device_(device),
       outputs_(std::move(outputs)) {
   future_ = c10::make_intrusive<c10::ivalue::Future>(
       c10::ListType::ofTensors(), std::vector<c10::Device>{device_});
@@ -207,6 +208,13 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
     cross_engine_ = std::make_unique<Engine>(topo_.node_id, topo_.n_nodes);
     hierarchical_ = true;
   }
+  if (env_flag("CGX_VERBOSE") && rank_ == 0) {
+    fprintf(stderr,
+            "[cgx] init: world=%d nodes=%d local_size=%d hierarchical=%d "
+            "device=%d\n",
+            size_, topo_.n_nodes, topo_.local_size, (int)hierarchical_,
+            device_index_);
+  }
 }
 
 template <typename Fn>
@@ -228,7 +236,8 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::collective(
     }
   }
   auto work = c10::make_intrusive<WorkCGX>(rank_, op, device,
-                                           std::move(outputs));
+                                           std::move(outputs),
+                                           c10d::opTypeToString(op).c_str());
   work->recordEnd(*stream_);
   return work;
 }
@@ -288,7 +297,8 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
                   s, device_index_));
     }
     auto work = c10::make_intrusive<WorkCGX>(rank_, c10d::OpType::ALLREDUCE,
-                                             t.device(), tensors);
+                                             t.device(), tensors,
+                                             "cgx:allreduce_compressed");
     work->recordEnd(fin_masq);
     return work;
   }

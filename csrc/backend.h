@@ -33,7 +33,8 @@ namespace cgx {
 class WorkCGX : public c10d::Work {
  public:
   WorkCGX(int rank, c10d::OpType op, at::Device device,
-          std::vector<at::Tensor> outputs);
+          std::vector<at::Tensor> outputs,
+          const char* profiling_title = "cgx");
   ~WorkCGX() override;
 
   // record the completion event on `stream` and mark the future completed
