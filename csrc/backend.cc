@@ -30,7 +30,10 @@ bool env_flag(const char* name) {
 
 void check_single(const std::vector<at::Tensor>& ts) {
   TORCH_CHECK(ts.size() == 1, "cgx: expected exactly one tensor per rank");
-  TORCH_CHECK(ts[0].is_contiguous(), "cgx: tensor must be contiguous");
+  // contiguity is only required on the RCCL/kernel path; the gloo CPU
+  // delegate handles arbitrary layouts itself
+  TORCH_CHECK(!ts[0].is_cuda() || ts[0].is_contiguous(),
+              "cgx: CUDA tensor must be contiguous");
 }
 
 }  // namespace
@@ -589,8 +592,24 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::alltoall(
   TORCH_CHECK((int)outputs.size() == size_ && (int)inputs.size() == size_);
   if (!inputs[0].is_cuda()) {
     TORCH_CHECK(cpu_, "cgx: no CPU delegate backend available");
-    c10d::AllToAllOptions o;
-    return cpu_->alltoall(outputs, inputs, o);
+    // gloo has no alltoall; compose it from ordered pairwise gloo p2p
+    // (deadlock-free: the lower rank of each pair sends first; the
+    // reference used MPI_Alltoall here, ProcessGroupCGX.cc:659)
+    outputs[rank_].copy_(inputs[rank_]);
+    for (int p = 0; p < size_; p++) {
+      if (p == rank_) continue;
+      std::vector<at::Tensor> st{inputs[p].contiguous()};
+      std::vector<at::Tensor> rt{outputs[p]};
+      if (rank_ < p) {
+        cpu_->send(st, p, 0)->wait();
+        cpu_->recv(rt, p, 0)->wait();
+      } else {
+        cpu_->recv(rt, p, 0)->wait();
+        cpu_->send(st, p, 0)->wait();
+      }
+    }
+    c10d::BarrierOptions bo;
+    return cpu_->barrier(bo);
   }
   std::vector<at::Tensor> ins = inputs, outs = outputs;
   return collective(outputs, inputs[0].device(), c10d::OpType::ALLTOALL,

@@ -111,6 +111,7 @@ EngineConfig EngineConfig::from_env() {
   c.stochastic = env_int("CGX_STOCHASTIC_ROUNDING", 1) != 0;
   c.skip_incomplete =
       env_int("CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS", 0) != 0;
+  c.dummy = env_int("CGX_DEBUG_DUMMY_COMPRESSION", 0) != 0;
   const char* fr = std::getenv("CGX_COMPRESSION_FAKE_RATIO");
   if (fr && *fr) c.fake_ratio = std::atof(fr);
   if (!(c.fake_ratio > 0.0 && c.fake_ratio <= 1.0)) c.fake_ratio = 1.0;
@@ -623,7 +624,7 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   std::vector<LayerView> comp_views;
   std::vector<std::pair<char*, int64_t>> uncomp;
   for (const auto& v : views) {
-    const bool c = v.bits <= 8 && v.numel > cfg.min_elems;
+    const bool c = !cfg.dummy && v.bits <= 8 && v.numel > cfg.min_elems;
     if (c) {
       comp_views.push_back(v);
     } else {

@@ -93,6 +93,7 @@ struct EngineConfig {
   bool ring = false;  // CGX_INNER_REDUCTION_TYPE=Ring (default SRA)
   double fake_ratio = 1.0;  // CGX_COMPRESSION_FAKE_RATIO (bandwidth expt)
   bool skip_incomplete = false;  // CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS
+  bool dummy = false;  // CGX_DEBUG_DUMMY_COMPRESSION: force uncompressed
   static EngineConfig from_env();  // re-read every bucket like the reference
 };
 

@@ -85,3 +85,43 @@ def _ws3(rank, ws):
 
 def test_allreduce_cpu_ws3():
     run_dist(_ws3, 3)
+
+
+def _subgroup(rank, ws):
+    g = dist.new_group([0, 1])
+    t = torch.full((8,), float(rank + 1))
+    if rank in (0, 1):
+        dist.all_reduce(t, group=g)
+        assert torch.equal(t, torch.full((8,), 3.0))
+    dist.barrier()
+
+
+def test_new_group_cpu():
+    run_dist(_subgroup, 2)
+
+
+def _more_collectives(rank, ws):
+    # gather
+    t = torch.full((4,), float(rank + 1))
+    out = [torch.zeros(4) for _ in range(ws)] if rank == 0 else None
+    dist.gather(t, out, dst=0)
+    if rank == 0:
+        for p in range(ws):
+            assert torch.equal(out[p], torch.full((4,), float(p + 1)))
+    # scatter
+    r = torch.zeros(4)
+    src = [torch.full((4,), float(p + 10)) for p in range(ws)] \
+        if rank == 0 else None
+    dist.scatter(r, src, src=0)
+    assert torch.equal(r, torch.full((4,), float(rank + 10)))
+    # alltoall
+    ins = list(torch.arange(float(ws * 2)).add(rank * 100).chunk(ws))
+    outs = list(torch.zeros(ws * 2).chunk(ws))
+    dist.all_to_all(outs, ins)
+    for p in range(ws):
+        assert torch.equal(outs[p], torch.arange(float(ws * 2))
+                           .add(p * 100).chunk(ws)[rank])
+
+
+def test_gather_scatter_alltoall_cpu():
+    run_dist(_more_collectives, 2)
