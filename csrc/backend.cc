@@ -549,10 +549,10 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::alltoall_base(
   at::Tensor in = input, out = output;
   std::vector<int64_t> osplit = outputSplitSizes, isplit = inputSplitSizes;
   if (osplit.empty()) {
-    TORCH_CHECK(out.numel() % size_ == 0 && in.numel() % size_ == 0);
-    osplit.assign(size_, out.numel() / size_ / (out.dim() ? 1 : 1));
-    isplit.assign(size_, in.numel() / size_);
+    TORCH_CHECK(out.numel() % size_ == 0 && in.numel() % size_ == 0,
+                "cgx: alltoall_base tensor not divisible by world size");
     osplit.assign(size_, out.numel() / size_);
+    isplit.assign(size_, in.numel() / size_);
   } else {
     // splits are in units of dim-0 rows
     int64_t orow = out.dim() > 0 && out.size(0) > 0 ? out.numel() / out.size(0) : 1;
