@@ -18,6 +18,11 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+# MIOpen's exhaustive kernel search can take minutes per fresh machine; FAST
+# find keeps startup bounded (override: CGX_BENCH_TUNE=1 enables full
+# benchmark-mode autotuning for maximum steady-state throughput).
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 import torch.distributed as dist
 import torch.nn as nn
@@ -51,7 +56,8 @@ def main():
     if use_cuda:
         assert torch.cuda.is_available(), "bench.py requires a GPU"
         torch.cuda.set_device(local_rank)
-        torch.backends.cudnn.benchmark = True
+        if os.environ.get("CGX_BENCH_TUNE", "0") == "1":
+            torch.backends.cudnn.benchmark = True
         device = torch.device("cuda", local_rank)
     else:
         device = torch.device("cpu")
