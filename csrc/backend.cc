@@ -220,12 +220,24 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
   }
 }
 
+namespace {
+void check_async_error(ncclComm_t comm) {
+  if (!comm) return;
+  ncclResult_t e = ncclSuccess;
+  (void)ncclCommGetAsyncError(comm, &e);
+  TORCH_CHECK(e == ncclSuccess,
+              "cgx: RCCL async error on communicator: ",
+              ncclGetErrorString(e));
+}
+}  // namespace
+
 template <typename Fn>
 c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::collective(
     std::vector<at::Tensor> outputs, at::Device device, c10d::OpType op,
     Fn&& fn) {
   std::lock_guard<std::mutex> lock(mu_);
   lazyInit(device);
+  check_async_error(comm_);
   c10::hip::HIPGuardMasqueradingAsCUDA dguard(device_index_);
   auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_index_);
   CGX_HIP_CHECK(hipEventRecord(start_ev_, cur.stream()));
@@ -265,6 +277,11 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
     // bucket's xGMI traffic.
     std::lock_guard<std::mutex> lock(mu_);
     lazyInit(t.device());
+    check_async_error(comm_);
+    if (hierarchical_) {
+      check_async_error(intra_comm_);
+      check_async_error(cross_comm_);
+    }
     c10::hip::HIPGuardMasqueradingAsCUDA dguard(device_index_);
     auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_index_);
     CGX_HIP_CHECK(hipEventRecord(start_ev_, cur.stream()));
