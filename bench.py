@@ -18,11 +18,6 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-# MIOpen's exhaustive kernel search can take minutes per fresh machine; FAST
-# find keeps startup bounded (override: CGX_BENCH_TUNE=1 enables full
-# benchmark-mode autotuning for maximum steady-state throughput).
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
-
 import torch
 import torch.distributed as dist
 import torch.nn as nn

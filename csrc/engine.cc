@@ -408,6 +408,56 @@ Engine::ChunkPlan Engine::plan(const std::vector<LayerView>& views,
   return pl;
 }
 
+void Engine::timer_begin(hipStream_t qs, bool enabled) {
+  timing_ = enabled;
+  timer_slot_ = -1;
+  if (!enabled) return;
+  PhaseTimer& t = timer_;
+  if (!t.inited) {
+    for (int r = 0; r < PhaseTimer::kRing; r++)
+      for (int e = 0; e < PhaseTimer::kEv; e++)
+        CGX_HIP_CHECK(hipEventCreate(&t.ev[r][e]));  // timing-capable
+    t.inited = true;
+  }
+  const int slot = t.cur;
+  t.cur = (t.cur + 1) % PhaseTimer::kRing;
+  if (t.pending[slot]) {
+    if (hipEventQuery(t.ev[slot][PhaseTimer::kEv - 1]) == hipSuccess) {
+      for (int p = 0; p + 1 < PhaseTimer::kEv; p++) {
+        float ms = 0.f;
+        if (hipEventElapsedTime(&ms, t.ev[slot][p], t.ev[slot][p + 1]) ==
+            hipSuccess)
+          t.sum_ms[p] += ms;
+      }
+      t.count++;
+      if (t.count % 50 == 0) {
+        fprintf(stderr,
+                "[cgx timings rank %d, %lld chunks] quantize %.3f ms | "
+                "comm1 %.3f | decode+requant %.3f | comm2 %.3f | decode2 "
+                "%.3f\n",
+                rank_, (long long)t.count, t.sum_ms[0] / t.count,
+                t.sum_ms[1] / t.count, t.sum_ms[2] / t.count,
+                t.sum_ms[3] / t.count, t.sum_ms[4] / t.count);
+      }
+      t.pending[slot] = false;
+    } else {
+      timing_ = false;  // ring full of in-flight chunks: skip this one
+      return;
+    }
+  }
+  timer_slot_ = slot;
+  CGX_HIP_CHECK(hipEventRecord(t.ev[slot][0], qs));
+}
+
+void Engine::timer_mark(int idx, hipStream_t stream) {
+  if (!timing_ || timer_slot_ < 0) return;
+  CGX_HIP_CHECK(hipEventRecord(timer_.ev[timer_slot_][idx], stream));
+}
+
+void Engine::timer_finish() {
+  if (timing_ && timer_slot_ >= 0) timer_.pending[timer_slot_] = true;
+}
+
 void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
                        ncclComm_t comm, hipStream_t qs,
                        const EngineConfig& cfg) {
@@ -460,6 +510,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
     }
     run_quantize(all, send1, dt, qs, cfg.stochastic);
   }
+  timer_mark(1, qs);
 
   // round 1 exchange on the comm stream (grouped p2p drives all xGMI links)
   chain(qs, comm_stream_);
@@ -474,6 +525,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
                               ncclUint8, p, comm, comm_stream_));
   }
   CGX_NCCL_CHECK(ncclGroupEnd());
+  timer_mark(2, comm_stream_);
 
   // decode-accumulate + self-quantize on the deq stream
   chain(comm_stream_, deq_stream_);
@@ -484,6 +536,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
     // through my own decode so all ranks end bit-identical
     run_quantize(rs[rank_], send2, dt, deq_stream_, cfg.stochastic);
   }
+  timer_mark(3, deq_stream_);
 
   // round 2 exchange on the comm stream
   chain(deq_stream_, comm_stream_);
@@ -498,6 +551,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
                               comm, comm_stream_));
   }
   CGX_NCCL_CHECK(ncclGroupEnd());
+  timer_mark(4, comm_stream_);
 
   // final decode: own chunk from send2, peers' chunks from recv2
   chain(comm_stream_, deq_stream_);
@@ -515,6 +569,8 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
     }
     run_dequant(all, recv2, 0, 1, /*add=*/false, dt, deq_stream_);
   }
+  timer_mark(5, deq_stream_);
+  timer_finish();
   CGX_HIP_CHECK(hipEventRecord(sslot.done_ev, deq_stream_));
   sslot.recorded = true;
 }
@@ -652,9 +708,11 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   const int64_t fusion_elems = std::max<int64_t>(256, cfg.fusion_bytes / es);
   std::vector<LayerView> cur;
   int64_t cur_n = 0;
+  static const bool timings_on = env_int("CGX_TIMINGS", 0) != 0;
   bool any_comp = false;
   auto run_chunk = [&](const std::vector<LayerView>& vs_in) {
     any_comp = true;
+    timer_begin(qs, timings_on && !cfg.ring);
     // CGX_COMPRESSION_FAKE_RATIO < 1: reduce only a fraction of each chunk
     // (bandwidth experiments; intentionally lossy -- reference
     // mpi_allreduce_operations.cc:143-144)

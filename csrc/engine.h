@@ -192,6 +192,27 @@ class Engine {
                    int64_t src_stride, int nsrc, bool add, DType dt,
                    hipStream_t stream);
 
+  // CGX_TIMINGS=1: per-phase GPU timing of each SRA chunk (quantize /
+  // round-1 comm / decode+requantize / round-2 comm / final decode),
+  // aggregated and printed every 50 chunks (the reference had no
+  // timers at all).
+  struct PhaseTimer {
+    static constexpr int kRing = 4;
+    static constexpr int kEv = 6;
+    hipEvent_t ev[kRing][kEv] = {};
+    bool pending[kRing] = {};
+    int cur = 0;
+    double sum_ms[kEv - 1] = {};
+    int64_t count = 0;
+    bool inited = false;
+  };
+  PhaseTimer timer_;
+  void timer_begin(hipStream_t qs, bool enabled);
+  void timer_mark(int idx, hipStream_t stream);  // idx 1..5
+  void timer_finish();
+  bool timing_ = false;
+  int timer_slot_ = -1;
+
   int rank_, size_;
   at::Tensor staging_;
   StagingSlot slots_[2];
