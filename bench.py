@@ -93,7 +93,7 @@ def main():
             opt.step()
             return loss
     else:
-        batch = args.batch or (32 if use_cuda else 2)
+        batch = args.batch or (64 if use_cuda else 2)
         seq = 128
         model = bert_large().to(device)
         data = torch.randint(0, 30522, (batch, seq), device=device)
