@@ -296,9 +296,7 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
         fin = intra_engine_->allreduce(t, intra_comm_, stream_->stream());
         if (topo_.local_rank == 0)
           fin = cross_engine_->allreduce(t, cross_comm_, fin);
-        CGX_NCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
-                                     nccl_dtype(t), /*root=*/0, intra_comm_,
-                                     fin));
+        fin = intra_engine_->broadcast(t, /*root=*/0, intra_comm_, fin);
       } else {
         fin = engine_->allreduce(t, comm_, stream_->stream());
       }

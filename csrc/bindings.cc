@@ -172,6 +172,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_quantization_bits", &cgx::py_set_bits);
   m.def("set_quantization_bucket_size", &cgx::py_set_bucket);
   m.def("clear_registry", [] { cgx::Registry::get().clear(); });
+  m.def("registry_snapshot", [] {
+    // [(bucket_idx, [numels], [(bits, bucket_size)])] in registration order
+    py::list out;
+    auto& reg = cgx::Registry::get();
+    for (const auto& b : reg.snapshot()) {
+      py::list numels, cfgs;
+      for (auto n : b.numels) numels.append(n);
+      for (const auto& c : b.cfgs)
+        cfgs.append(py::make_tuple(c.bits, c.bucket_size));
+      out.append(py::make_tuple(b.idx, numels, cfgs));
+    }
+    return out;
+  });
 
   m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
         py::arg("bucket_size"), py::arg("stochastic") = false,

@@ -75,6 +75,11 @@ bool Registry::empty() {
   return order_.empty();
 }
 
+std::vector<Registry::BucketInfo> Registry::snapshot() {
+  std::lock_guard<std::mutex> g(mu_);
+  return order_;
+}
+
 bool Registry::next(int64_t numel, BucketInfo* out) {
   std::lock_guard<std::mutex> g(mu_);
   if (order_.empty()) return false;
@@ -112,6 +117,7 @@ EngineConfig EngineConfig::from_env() {
   c.skip_incomplete =
       env_int("CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS", 0) != 0;
   c.dummy = env_int("CGX_DEBUG_DUMMY_COMPRESSION", 0) != 0;
+  c.intra_compress = env_int("CGX_INTRA_COMPRESS", 1) != 0;
   const char* fr = std::getenv("CGX_COMPRESSION_FAKE_RATIO");
   if (fr && *fr) c.fake_ratio = std::atof(fr);
   if (!(c.fake_ratio > 0.0 && c.fake_ratio <= 1.0)) c.fake_ratio = 1.0;

@@ -68,6 +68,7 @@ class Registry {
   // back to whole-tensor default config instead of corrupting the mapping).
   // Returns a copy (thread safety) or nullopt-like empty BucketInfo.
   bool next(int64_t numel, BucketInfo* out);
+  std::vector<BucketInfo> snapshot();
 
  private:
   std::mutex mu_;
@@ -94,6 +95,7 @@ struct EngineConfig {
   double fake_ratio = 1.0;  // CGX_COMPRESSION_FAKE_RATIO (bandwidth expt)
   bool skip_incomplete = false;  // CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS
   bool dummy = false;  // CGX_DEBUG_DUMMY_COMPRESSION: force uncompressed
+  bool intra_compress = true;  // CGX_INTRA_COMPRESS
   static EngineConfig from_env();  // re-read every bucket like the reference
 };
 
@@ -140,6 +142,15 @@ class Engine {
 
   hipStream_t comm_stream() const { return comm_stream_; }
   hipStream_t deq_stream() const { return deq_stream_; }
+
+  // Compressed broadcast (reference Reducer::Broadcast parity,
+  // reducer.cc:96-160): root quantizes with the default env config, the
+  // compressed bytes travel once over the wire, everyone decodes.  Falls
+  // back to plain ncclBroadcast when compression is off.  Used by the
+  // hierarchical intra-node broadcast when CGX_INTRA_COMPRESS=1 (default,
+  // matching the reference, mpi_allreduce_operations.cc:134).
+  hipStream_t broadcast(at::Tensor t, int root, ncclComm_t comm,
+                        hipStream_t qs);
 
   // Mirror of the partition walk (reference Quantizer::GetSizesAndOffsets,
   // compressor.cc:265-299); exposed for tests via bindings.

@@ -75,3 +75,33 @@ def _compat_shim_worker(rank, ws):
 
 def test_reference_import_compat():
     run_dist(_compat_shim_worker, 2)
+
+
+def _registry_snapshot_worker(rank, ws):
+    import torch_cgx_amd
+    from torch_cgx_amd import _C
+    _C.clear_registry()
+    state = torch_cgx_amd.CGXState(None, layer_min_size=32,
+                                   compression_params={"bits": 4,
+                                                       "bucket_size": 128})
+    _train(Net(), 4, rank, torch_cgx_amd.cgx_hook, state)
+    snap = _C.registry_snapshot()
+    assert len(snap) >= 1  # DDP registered at least one bucket
+    total = sum(sum(numels) for (_idx, numels, _cfgs) in snap)
+    n_params = sum(p.numel() for p in Net().parameters())
+    assert total == n_params, (total, n_params)
+    # layerwise filter: 1-D tensors (biases) registered with bits=32
+    saw_fp, saw_q = False, False
+    for (_idx, numels, cfgs) in snap:
+        for n, (bits, bucket) in zip(numels, cfgs):
+            if bits == 32:
+                saw_fp = True
+            else:
+                assert bits == 4 and bucket == 128
+                saw_q = True
+    assert saw_fp and saw_q
+    _C.clear_registry()
+
+
+def test_registry_snapshot_after_hook():
+    run_dist(_registry_snapshot_worker, 2)
