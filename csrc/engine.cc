@@ -657,6 +657,39 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
   run_dequant(all, segs, 0, 1, /*add=*/false, dt, stream);
 }
 
+hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
+                              hipStream_t qs) {
+  const EngineConfig cfg = EngineConfig::from_env();
+  const int64_t n = t.numel();
+  const bool compress = !cfg.dummy && cfg.intra_compress &&
+                        cfg.default_bits <= 8 && n > cfg.min_elems &&
+                        (t.scalar_type() == at::kFloat ||
+                         t.scalar_type() == at::kHalf ||
+                         t.scalar_type() == at::kBFloat16);
+  if (!compress || size_ <= 1) {
+    if (size_ > 1) {
+      CGX_NCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), n,
+                                   nccl_dtype(t), root, comm, qs));
+    }
+    return qs;
+  }
+  const DType dt = dtype_of(t);
+  const int64_t bytes =
+      buffer_size(n, dt, cfg.default_bits, cfg.default_bucket,
+                  cfg.skip_incomplete);
+  uint8_t* buf = staging(bytes);
+  std::vector<Slice> sl{Slice{static_cast<char*>(t.data_ptr()), n,
+                              cfg.default_bits, cfg.default_bucket, 0,
+                              cfg.skip_incomplete}};
+  if (rank_ == root) {
+    run_quantize(sl, buf, dt, qs, cfg.stochastic);
+  }
+  CGX_NCCL_CHECK(ncclBroadcast(buf, buf, bytes, ncclUint8, root, comm, qs));
+  // every rank (root included) decodes the same bytes -> bit-identical
+  run_dequant(sl, buf, 0, 1, /*add=*/false, dt, qs);
+  return qs;
+}
+
 hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
                               hipStream_t qs) {
   if (size_ <= 1) return qs;
