@@ -681,6 +681,18 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::recv(
                     });
 }
 
+c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::recvAnysource(
+    std::vector<at::Tensor>& tensors, int tag) {
+  check_single(tensors);
+  if (!tensors[0].is_cuda()) {
+    TORCH_CHECK(cpu_, "cgx: no CPU delegate backend available");
+    return cpu_->recvAnysource(tensors, tag);
+  }
+  TORCH_CHECK(false,
+              "cgx: recvAnysource is not supported for GPU tensors (RCCL "
+              "p2p requires a known source rank)");
+}
+
 c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::barrier(
     const c10d::BarrierOptions& opts) {
   // Host-level rendezvous; if the GPU side is active, drain the side stream

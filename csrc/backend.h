@@ -122,6 +122,8 @@ class ProcessGroupCGX : public c10d::Backend {
                                       int dstRank, int tag) override;
   c10::intrusive_ptr<c10d::Work> recv(std::vector<at::Tensor>& tensors,
                                       int srcRank, int tag) override;
+  c10::intrusive_ptr<c10d::Work> recvAnysource(
+      std::vector<at::Tensor>& tensors, int tag) override;
   c10::intrusive_ptr<c10d::Work> barrier(
       const c10d::BarrierOptions& opts) override;
 

@@ -100,3 +100,28 @@ def test_ring_error_bound(ws, bits):
         assert err < bound, (err, bound)
     for o in out[1:]:
         assert torch.equal(o, out[0])
+
+
+def test_fuzz_random_configs():
+    """Randomized layers/configs/world sizes: SRA result within an analytic
+    bound of the exact sum, all ranks bitwise identical."""
+    rng = np.random.default_rng(7)
+    for trial in range(10):
+        ws = int(rng.choice([2, 3, 4, 8]))
+        nl = int(rng.integers(1, 5))
+        layers = [int(rng.integers(1, 4000)) for _ in range(nl)]
+        cfgs = [(int(rng.choice([2, 4, 8])), int(rng.choice([64, 512, 1024])))
+                for _ in range(nl)]
+        n = sum(layers)
+        tensors = [torch.randn(n) for _ in range(ws)]
+        out = sra_sim.sra_allreduce(tensors, layers, cfgs)
+        exact = sum(tensors)
+        # per-element bound: 2 * unit-scale * ws * (ws+1); use the max
+        # range across ranks as the scale proxy
+        scale = max(t.abs().max().item() for t in tensors) * 2
+        worst_bits = min(b for b, _ in cfgs)
+        bound = 2 * scale / ((1 << worst_bits) - 1) * ws * (ws + 1)
+        err = (out[0] - exact).abs().max().item()
+        assert err < bound, (trial, ws, layers, cfgs, err, bound)
+        for o in out[1:]:
+            assert torch.equal(o, out[0])
