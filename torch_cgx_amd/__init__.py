@@ -36,8 +36,6 @@ except ImportError as e:  # pragma: no cover
 
 __version__ = "0.1.0"
 
-_TIMEOUT_DEFAULT = None
-
 
 def _create_cgx(store, rank, size, timeout):
     """Backend creator registered with torch.distributed for name 'cgx'."""

@@ -136,7 +136,6 @@ def unpack_levels(packed: np.ndarray, n: int, bits: int) -> np.ndarray:
     """Inverse of pack_levels -> uint8 levels of length n."""
     ngroups = (n + PACK_SIZE - 1) // PACK_SIZE
     by = np.zeros((ngroups, 8), dtype=np.uint8)
-    flat = by.reshape(-1)
     take = min(len(packed), ngroups * bits)
     # bytes of group i live at packed[i*bits : i*bits+bits]
     src = packed[:take]
