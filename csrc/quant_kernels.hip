@@ -20,6 +20,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <type_traits>
 
 #include "compress.h"
 
@@ -138,6 +139,48 @@ template <>
 __device__ __forceinline__ uint32_t f2raw<__hip_bfloat16>(float f) {
   return bitcast<uint16_t>(__float2bfloat16(f));
 }
+
+// Packed-pair min/max for 16-bit dtypes (v_pk_min/max_f16|bf16): phase A of
+// the fp16/bf16 quantize was 87% VALU-busy with per-element convert+min+max;
+// packed ops process 2 elements per instruction without unpacking.
+template <typename T>
+struct Pk2;
+template <>
+struct Pk2<__half> {
+  using P = _Float16 __attribute__((ext_vector_type(2)));
+  static constexpr uint32_t kInf = 0x7C007C00u;   // +inf, +inf
+  static constexpr uint32_t kNInf = 0xFC00FC00u;  // -inf, -inf
+  static __device__ __forceinline__ P min(P a, P b) {
+    return __builtin_elementwise_min(a, b);
+  }
+  static __device__ __forceinline__ P max(P a, P b) {
+    return __builtin_elementwise_max(a, b);
+  }
+  static __device__ __forceinline__ float lo(P v) {
+    return static_cast<float>(v[0]);
+  }
+  static __device__ __forceinline__ float hi(P v) {
+    return static_cast<float>(v[1]);
+  }
+};
+template <>
+struct Pk2<__hip_bfloat16> {
+  using P = __bf16 __attribute__((ext_vector_type(2)));
+  static constexpr uint32_t kInf = 0x7F807F80u;
+  static constexpr uint32_t kNInf = 0xFF80FF80u;
+  static __device__ __forceinline__ P min(P a, P b) {
+    return __builtin_elementwise_min(a, b);
+  }
+  static __device__ __forceinline__ P max(P a, P b) {
+    return __builtin_elementwise_max(a, b);
+  }
+  static __device__ __forceinline__ float lo(P v) {
+    return static_cast<float>(v[0]);
+  }
+  static __device__ __forceinline__ float hi(P v) {
+    return static_cast<float>(v[1]);
+  }
+};
 
 // 8-element vector load/store (one pack).  aligned16 is wave-uniform for the
 // quantize kernel (slice bases are pack-aligned) and per-thread for dequant.
@@ -307,13 +350,37 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     // fast path: full bucket, whole groups per lane (wave-uniform branch)
     const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
     if (full && ngroups <= MAXG * kWave) {
-      for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
-        load8<T>(in + g * 8, true, stash[gi]);
+      if constexpr (sizeof(T) == 2) {
+        using PK = Pk2<T>;
+        typename PK::P pmin = bitcast<typename PK::P>(PK::kInf);
+        typename PK::P pmax = bitcast<typename PK::P>(PK::kNInf);
+        for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+          const int4 a = *reinterpret_cast<const int4*>(in + g * 8);
+          const uint32_t w[4] = {static_cast<uint32_t>(a.x),
+                                 static_cast<uint32_t>(a.y),
+                                 static_cast<uint32_t>(a.z),
+                                 static_cast<uint32_t>(a.w)};
 #pragma unroll
-        for (int j = 0; j < 8; j++) {
-          const float f = raw2f<T>(stash[gi][j]);
-          lmin = fminf(lmin, f);
-          lmax = fmaxf(lmax, f);
+          for (int k = 0; k < 4; k++) {
+            const auto v = bitcast<typename PK::P>(w[k]);
+            pmin = PK::min(pmin, v);
+            pmax = PK::max(pmax, v);
+          }
+#pragma unroll
+          for (int j = 0; j < 8; j++)
+            stash[gi][j] = (w[j >> 1] >> ((j & 1) * 16)) & 0xFFFF;
+        }
+        lmin = fminf(PK::lo(pmin), PK::hi(pmin));
+        lmax = fmaxf(PK::lo(pmax), PK::hi(pmax));
+      } else {
+        for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+          load8<T>(in + g * 8, true, stash[gi]);
+#pragma unroll
+          for (int j = 0; j < 8; j++) {
+            const float f = raw2f<T>(stash[gi][j]);
+            lmin = fminf(lmin, f);
+            lmax = fmaxf(lmax, f);
+          }
         }
       }
     } else {
@@ -356,23 +423,27 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
       const bool live = unitf >= kEps;
       const float rinv = 1.0f / unitf;  // hoisted: fp32 div is 1/4 VALU rate
       if (full && ngroups <= MAXG * kWave && live) {
+        // 8 levels fit a uint32 when BITS <= 4: halves the shift/or cost
+        using acc_t =
+            typename std::conditional<(BITS <= 4), uint32_t, uint64_t>::type;
         for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
           const uint64_t pr =
               stochastic
                   ? rand_pack(seed, (static_cast<uint64_t>(lo) << 44) |
                                         static_cast<uint64_t>(gbase + g))
                   : 0;
-          uint64_t value = 0;
+          acc_t value = 0;
 #pragma unroll
           for (int j = 0; j < 8; j++) {
             const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
             const float dd = (raw2f<T>(stash[gi][j]) - minf) * rinv + rnd;
             const uint32_t level =
                 static_cast<uint32_t>(fminf(floorf(dd), divisor));
-            value |= static_cast<uint64_t>(level & ((1u << BITS) - 1))
+            value |= static_cast<acc_t>(level & ((1u << BITS) - 1))
                      << (j * BITS);
           }
-          store_bytes(packed + (gbase + g) * BITS, value, BITS);
+          store_bytes(packed + (gbase + g) * BITS,
+                      static_cast<uint64_t>(value), BITS);
         }
         continue;
       }
