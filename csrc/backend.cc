@@ -292,10 +292,15 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
       const bool hier = hierarchical_ && !(ib && std::strcmp(ib, "0") == 0);
       if (hier) {
         // intra-node compressed allreduce over xGMI, cross-node reduction on
-        // node leaders, then intra broadcast of the result
-        fin = intra_engine_->allreduce(t, intra_comm_, stream_->stream());
+        // node leaders, then intra broadcast of the result.  The layer
+        // registry is consulted exactly ONCE per bucket so the cursor stays
+        // in lockstep on every rank (leaders run two engine passes).
+        cgx::Registry::BucketInfo info;
+        const bool matched = cgx::Registry::get().next(t.numel(), &info);
+        fin = intra_engine_->allreduce(t, intra_comm_, stream_->stream(),
+                                       &info, matched);
         if (topo_.local_rank == 0)
-          fin = cross_engine_->allreduce(t, cross_comm_, fin);
+          fin = cross_engine_->allreduce(t, cross_comm_, fin, &info, matched);
         fin = intra_engine_->broadcast(t, /*root=*/0, intra_comm_, fin);
       } else {
         fin = engine_->allreduce(t, comm_, stream_->stream());

@@ -691,7 +691,9 @@ hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
 }
 
 hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
-                              hipStream_t qs) {
+                              hipStream_t qs,
+                              const Registry::BucketInfo* forced,
+                              bool forced_match) {
   if (size_ <= 1) return qs;
   TORCH_CHECK(bucket.is_contiguous(), "cgx: bucket must be contiguous");
   const EngineConfig cfg = EngineConfig::from_env();
@@ -701,7 +703,13 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   const int64_t numel = bucket.numel();
 
   Registry::BucketInfo info;
-  const bool matched = Registry::get().next(numel, &info);
+  bool matched;
+  if (forced) {
+    matched = forced_match;
+    if (matched) info = *forced;
+  } else {
+    matched = Registry::get().next(numel, &info);
+  }
   std::vector<LayerView> views;
   if (matched) {
     int64_t off = 0;

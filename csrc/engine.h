@@ -138,7 +138,13 @@ class Engine {
   // two, the NEXT bucket's quantize overlaps this bucket's xGMI traffic
   // (the reference serialized everything on one side stream).  Returns the
   // stream carrying the final operation (record the Work end event there).
-  hipStream_t allreduce(at::Tensor bucket, ncclComm_t comm, hipStream_t qs);
+  // `forced`: pre-resolved layer info (hierarchical mode resolves the
+  // registry ONCE per bucket and passes it to both the intra and cross
+  // engines -- each engine consuming a registry cursor step would
+  // desynchronize leader vs non-leader ranks when bucket totals repeat).
+  hipStream_t allreduce(at::Tensor bucket, ncclComm_t comm, hipStream_t qs,
+                        const Registry::BucketInfo* forced = nullptr,
+                        bool forced_match = false);
 
   hipStream_t comm_stream() const { return comm_stream_; }
   hipStream_t deq_stream() const { return deq_stream_; }
