@@ -269,9 +269,21 @@ void Engine::partition(int64_t num_elements, int ws,
   }
 }
 
-uint8_t* Engine::staging(int64_t bytes) {
+uint8_t* Engine::staging(int64_t bytes, hipStream_t user) {
   if (bytes <= 0) bytes = 1;
   if (!staging_.defined() || staging_.numel() < bytes) {
+    if (staging_.defined()) {
+      // the old buffer may still be read by queued work on the user/comm/deq
+      // streams; let the caching allocator defer reuse past those streams
+      for (hipStream_t s : {user, comm_stream_, deq_stream_}) {
+        if (!s) continue;
+        c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
+            recordStreamMasqueradingAsCUDA(
+                staging_.storage().data_ptr(),
+                c10::hip::getStreamFromExternalMasqueradingAsCUDA(
+                    s, staging_.device().index()));
+      }
+    }
     int64_t cap = 1 << 20;
     while (cap < bytes) cap *= 2;
     staging_ = at::empty({cap}, at::TensorOptions()
@@ -603,7 +615,7 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
     seg_total += comp[k];
     maxc = std::max(maxc, comp[k]);
   }
-  uint8_t* base = staging(seg_total + 2 * maxc);
+  uint8_t* base = staging(seg_total + 2 * maxc, stream);
   uint8_t* segs = base;
   uint8_t* tmp_send = base + seg_total;
   uint8_t* tmp_recv = tmp_send + maxc;
@@ -677,7 +689,7 @@ hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
   const int64_t bytes =
       buffer_size(n, dt, cfg.default_bits, cfg.default_bucket,
                   cfg.skip_incomplete);
-  uint8_t* buf = staging(bytes);
+  uint8_t* buf = staging(bytes, qs);
   std::vector<Slice> sl{Slice{static_cast<char*>(t.data_ptr()), n,
                               cfg.default_bits, cfg.default_bucket, 0,
                               cfg.skip_incomplete}};

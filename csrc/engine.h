@@ -197,7 +197,7 @@ class Engine {
   void ring_chunk(const std::vector<LayerView>& views, DType dt,
                   ncclComm_t comm, hipStream_t stream,
                   const EngineConfig& cfg);
-  uint8_t* staging(int64_t bytes);                       // single-stream path
+  uint8_t* staging(int64_t bytes, hipStream_t user);     // single-stream path
   uint8_t* slot_bytes(StagingSlot& slot, int64_t bytes); // pipelined path
   void chain(hipStream_t from, hipStream_t to);  // event: `to` waits `from`
   hipEvent_t next_ev();
