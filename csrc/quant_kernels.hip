@@ -325,7 +325,7 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
       static_cast<int64_t>(blockIdx.x) * (kThreads / kWave) + (threadIdx.x / kWave);
   const int64_t nw = static_cast<int64_t>(gridDim.x) * (kThreads / kWave);
   constexpr float divisor = static_cast<float>((1 << BITS) - 1);
-  constexpr int MAXG = 2;  // register-stashed packs per lane (bucket <= 1024)
+  constexpr int MAXG = 4;  // register-stashed packs per lane (bucket <= 2048)
 
   for (int64_t b = wid; b < total_buckets; b += nw) {
     int lo = 0, hi = nslices;
