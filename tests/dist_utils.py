@@ -31,10 +31,24 @@ def _worker(rank, world_size, port, fn, args, backend):
 
 def run_dist(fn, world_size=2, backend="cgx", args=()):
     """Run fn(rank, world_size, *args) on world_size processes."""
-    port = free_port()
     # spawn, not fork: earlier tests warm torch's OpenMP pool in the pytest
     # parent, and forking a threaded parent deadlocks DDP workers.
     start = os.environ.get("CGX_TEST_START_METHOD", "spawn")
-    mp.start_processes(_worker,
-                       args=(world_size, port, fn, args, backend),
-                       nprocs=world_size, join=True, start_method=start)
+    last = None
+    for attempt in range(2):  # one retry shields against TCP port races
+        port = free_port()
+        try:
+            mp.start_processes(_worker,
+                               args=(world_size, port, fn, args, backend),
+                               nprocs=world_size, join=True,
+                               start_method=start)
+            return
+        except Exception as e:  # pragma: no cover - flake path
+            last = e
+            msg = str(e)
+            transient = any(k in msg for k in
+                            ("Address already in use", "bind", "ECONNREFUSED",
+                             "Connection reset", "timed out"))
+            if not transient or attempt == 1:
+                raise
+    raise last
