@@ -242,6 +242,18 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::collective(
   auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_index_);
   CGX_HIP_CHECK(hipEventRecord(start_ev_, cur.stream()));
   CGX_HIP_CHECK(hipStreamWaitEvent(stream_->stream(), start_ev_, 0));
+  // RCCL forbids unordered concurrent ops on one communicator: fence this
+  // passthrough op behind any compressed-allreduce traffic still queued on
+  // the engine's comm/decode streams (the compressed path itself chains the
+  // other direction through start_ev_, so ordering is total).
+  for (Engine* e : {engine_.get(), intra_engine_.get(),
+                    cross_engine_.get()}) {
+    if (!e) continue;
+    for (hipStream_t s : {e->comm_stream(), e->deq_stream()}) {
+      CGX_HIP_CHECK(hipEventRecord(start_ev_, s));
+      CGX_HIP_CHECK(hipStreamWaitEvent(stream_->stream(), start_ev_, 0));
+    }
+  }
   c10::hip::HIPStreamGuardMasqueradingAsCUDA sguard(stream_->unwrap());
   fn(stream_->stream());
   for (const auto& t : outputs) {
