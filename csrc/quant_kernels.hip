@@ -349,79 +349,6 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     float lmin = INFINITY, lmax = -INFINITY;
     // fast path: full bucket, whole groups per lane (wave-uniform branch)
     const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
-
-    if constexpr (sizeof(T) == 4) {
-      // fp32 transposed layout: one float4 per lane per step -- fully
-      // coalesced 16B/lane loads (the 8-float-per-lane layout strides 32B
-      // and halves load efficiency).  Each lane encodes 4 levels; adjacent
-      // lanes merge their halves of a pack with one DPP quad-perm swap.
-      constexpr int kMaxSubs = 2 * MAXG;
-      const int nsubs = cur >> 2;
-      if (full && nsubs <= kMaxSubs * kWave) {
-        uint32_t stash4[kMaxSubs][4];
-        int nwi = 0;
-        for (int w = lane, wi = 0; w < nsubs; w += kWave, wi++, nwi++) {
-          const int4 a = *reinterpret_cast<const int4*>(in + w * 4);
-          stash4[wi][0] = a.x; stash4[wi][1] = a.y;
-          stash4[wi][2] = a.z; stash4[wi][3] = a.w;
-#pragma unroll
-          for (int j = 0; j < 4; j++) {
-            const float f = raw2f<T>(stash4[wi][j]);
-            lmin = fminf(lmin, f);
-            lmax = fmaxf(lmax, f);
-          }
-        }
-        wave_minmax(lmin, lmax);
-        const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
-        const float unitf = raw2f<T>(unit_raw);
-        const float minf = lmin;
-        R* meta = reinterpret_cast<R*>(d.out);
-        if (lane == 0) {
-          meta[2 * lb] = static_cast<R>(unit_raw);
-          meta[2 * lb + 1] = static_cast<R>(f2raw<T>(lmin));
-        }
-        if constexpr (ENCODE) {
-          uint8_t* packed =
-              reinterpret_cast<uint8_t*>(d.out) + 2 * sizeof(R) * nb_slice;
-          const int64_t gbase = bstart >> 3;
-          const bool live = unitf >= kEps;
-          const float rinv = 1.0f / unitf;
-          const int h = lane & 1;  // which half of the pack this lane holds
-          for (int w = lane, wi = 0; wi < nwi; w += kWave, wi++) {
-            const int64_t g = w >> 1;
-            uint32_t partial = 0;
-            if (live) {
-              const uint64_t pr =
-                  stochastic
-                      ? rand_pack(seed, (static_cast<uint64_t>(lo) << 44) |
-                                            static_cast<uint64_t>(gbase + g))
-                      : 0;
-#pragma unroll
-              for (int j = 0; j < 4; j++) {
-                const float rnd =
-                    stochastic ? rand_lane(pr, j + 4 * h) : 0.5f;
-                const float dd =
-                    (raw2f<T>(stash4[wi][j]) - minf) * rinv + rnd;
-                const uint32_t level =
-                    static_cast<uint32_t>(fminf(floorf(dd), divisor));
-                partial |= (level & ((1u << BITS) - 1)) << (j * BITS);
-              }
-            }
-            const uint32_t swapped = static_cast<uint32_t>(
-                __builtin_amdgcn_update_dpp(0, static_cast<int>(partial),
-                                            0xB1, 0xf, 0xf, true));
-            if (h == 0) {
-              const uint64_t value =
-                  static_cast<uint64_t>(partial) |
-                  (static_cast<uint64_t>(swapped) << (4 * BITS));
-              store_bytes(packed + (gbase + g) * BITS, value, BITS);
-            }
-          }
-        }
-        continue;
-      }
-    }
-
     if (full && ngroups <= MAXG * kWave) {
       if constexpr (sizeof(T) == 2) {
         using PK = Pk2<T>;
