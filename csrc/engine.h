@@ -111,7 +111,7 @@ class DescRing {
   void* commit(size_t bytes, hipStream_t stream);
 
  private:
-  static constexpr int kSlots = 8;
+  static constexpr int kSlots = 16;
   struct Slot {
     void* host = nullptr;
     void* dev = nullptr;
@@ -245,6 +245,5 @@ class Engine {
 
 DType dtype_of(const at::Tensor& t);
 ncclDataType_t nccl_dtype(const at::Tensor& t);
-ncclRedOp_t nccl_op(int reduce_op_enum);  // c10d::ReduceOp::RedOpType value
 
 }  // namespace cgx
