@@ -125,3 +125,24 @@ def _more_collectives(rank, ws):
 
 def test_gather_scatter_alltoall_cpu():
     run_dist(_more_collectives, 2)
+
+
+def _base_variants(rank, ws):
+    # all_gather_into_tensor (_allgather_base delegation)
+    t = torch.full((4,), float(rank + 1))
+    out = torch.zeros(4 * ws)
+    dist.all_gather_into_tensor(out, t)
+    for p in range(ws):
+        assert torch.equal(out[p * 4:(p + 1) * 4],
+                           torch.full((4,), float(p + 1)))
+    # alltoall_base equal splits
+    inp = torch.arange(float(ws * 3)) + rank * 100
+    outp = torch.zeros(ws * 3)
+    dist.all_to_all_single(outp, inp)
+    for p in range(ws):
+        expect = torch.arange(float(ws * 3))[rank * 3:(rank + 1) * 3] + p * 100
+        assert torch.equal(outp[p * 3:(p + 1) * 3], expect)
+
+
+def test_base_collectives_cpu():
+    run_dist(_base_variants, 2)
