@@ -321,9 +321,14 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
     auto fin_masq = c10::hip::getStreamFromExternalMasqueradingAsCUDA(
         fin, device_index_);
     c10::hip::HIPStreamGuardMasqueradingAsCUDA sguard(fin_masq.unwrap());
-    for (hipStream_t s : {stream_->stream(), engine_ ? engine_->comm_stream()
-                                                     : nullptr,
-                          engine_ ? engine_->deq_stream() : nullptr}) {
+    std::vector<hipStream_t> touched{stream_->stream()};
+    for (Engine* e : {engine_.get(), intra_engine_.get(),
+                      cross_engine_.get()}) {
+      if (!e) continue;
+      touched.push_back(e->comm_stream());
+      touched.push_back(e->deq_stream());
+    }
+    for (hipStream_t s : touched) {
       if (!s) continue;
       c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
           recordStreamMasqueradingAsCUDA(
