@@ -133,6 +133,14 @@ ProcessGroupCGX::ProcessGroupCGX(c10::intrusive_ptr<c10d::Store> store,
       cpu_(std::move(cpu_delegate)) {}
 
 ProcessGroupCGX::~ProcessGroupCGX() {
+  // drain pending work before tearing communicators down
+  if (stream_) (void)hipStreamSynchronize(stream_->stream());
+  for (Engine* e : {engine_.get(), intra_engine_.get(),
+                    cross_engine_.get()}) {
+    if (!e) continue;
+    (void)hipStreamSynchronize(e->comm_stream());
+    (void)hipStreamSynchronize(e->deq_stream());
+  }
   if (intra_comm_) (void)ncclCommDestroy(intra_comm_);
   if (cross_comm_) (void)ncclCommDestroy(cross_comm_);
   if (comm_) (void)ncclCommDestroy(comm_);
