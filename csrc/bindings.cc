@@ -22,10 +22,21 @@ DType dtype_arg(const at::Tensor& t) { return dtype_of(t); }
 // Compress a 1-D CUDA tensor; returns the uint8 compressed buffer
 // (zero-filled alignment padding so byte comparisons are well-defined).
 at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
-                       bool stochastic, int64_t seed, bool skip_incomplete) {
+                       bool stochastic, int64_t seed, bool skip_incomplete,
+                       c10::optional<at::Tensor> feedback) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "quantize: CUDA contiguous");
   TORCH_CHECK(bits >= 1 && bits <= 8, "quantize: bits in [1,8]");
   TORCH_CHECK(bucket_size >= 1, "quantize: bucket_size >= 1");
+  void* fbp = nullptr;
+  if (feedback.has_value()) {
+    TORCH_CHECK(feedback->is_cuda() && feedback->is_contiguous() &&
+                    feedback->numel() == x.numel() &&
+                    feedback->scalar_type() == x.scalar_type(),
+                "quantize: feedback must match x");
+    TORCH_CHECK(bucket_size % 8 == 0,
+                "quantize: error feedback requires bucket_size % 8 == 0");
+    fbp = feedback->data_ptr();
+  }
   const DType dt = dtype_arg(x);
   const int64_t n = x.numel();
   const int64_t bytes =
@@ -38,7 +49,7 @@ at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
     QuantDesc d;
     int64_t cum[2];
   } hb;
-  hb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), n,
+  hb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), fbp, n,
                    (int32_t)bucket_size,
                    skip_incomplete ? kFlagSkipIncomplete : 0};
   hb.cum[0] = 0;
@@ -188,7 +199,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
         py::arg("bucket_size"), py::arg("stochastic") = false,
-        py::arg("seed") = 0, py::arg("skip_incomplete") = false);
+        py::arg("seed") = 0, py::arg("skip_incomplete") = false,
+        py::arg("feedback") = py::none());
   m.def("dequantize", &cgx::py_dequantize, py::arg("comp"), py::arg("out"),
         py::arg("bits"), py::arg("bucket_size"), py::arg("add") = false,
         py::arg("skip_incomplete") = false);

@@ -40,9 +40,14 @@ inline int64_t buffer_size(int64_t n, DType dt, int bits, int bucket_size,
 constexpr int32_t kFlagSkipIncomplete = 1;
 
 // One quantize work item: compress `n` elems at `in` into `out` bytes.
+// fb (optional): error-feedback residual of the same length/dtype as `in`;
+// the encoded value is in[i]+fb[i] (rounded to T) and fb is updated in place
+// to the new residual.  Wire format is unchanged.  (The reference carried
+// this plumbing but never enabled it, cuda_compression_operations.cu:713.)
 struct QuantDesc {
   const void* in;
   uint8_t* out;
+  void* fb;
   int64_t n;
   int32_t bucket;
   int32_t flags;  // kFlagSkipIncomplete

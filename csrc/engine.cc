@@ -118,6 +118,7 @@ EngineConfig EngineConfig::from_env() {
       env_int("CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS", 0) != 0;
   c.dummy = env_int("CGX_DEBUG_DUMMY_COMPRESSION", 0) != 0;
   c.intra_compress = env_int("CGX_INTRA_COMPRESS", 1) != 0;
+  c.error_feedback = env_int("CGX_ERROR_FEEDBACK", 0) != 0;
   const char* fr = std::getenv("CGX_COMPRESSION_FAKE_RATIO");
   if (fr && *fr) c.fake_ratio = std::atof(fr);
   if (!(c.fake_ratio > 0.0 && c.fake_ratio <= 1.0)) c.fake_ratio = 1.0;
@@ -318,8 +319,21 @@ uint8_t* Engine::slot_bytes(StagingSlot& slot, int64_t bytes) {
   return slot.buf.data_ptr<uint8_t>();
 }
 
+at::Tensor& Engine::feedback_buf(const void* key_ptr, int64_t numel,
+                                 int phase, at::ScalarType st) {
+  auto key = std::make_tuple(key_ptr, numel, phase);
+  auto it = fb_bufs_.find(key);
+  if (it == fb_bufs_.end()) {
+    auto t = at::zeros({numel}, at::TensorOptions().dtype(st).device(
+                                    at::kCUDA));
+    it = fb_bufs_.emplace(key, std::move(t)).first;
+  }
+  return it->second;
+}
+
 void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
-                          DType dt, hipStream_t stream, bool stochastic) {
+                          DType dt, hipStream_t stream, bool stochastic,
+                          char* fb_base) {
   // group by (bits, bucket%8==0)
   std::map<std::pair<int, bool>, std::vector<const Slice*>> groups;
   for (const auto& s : slices) {
@@ -338,7 +352,11 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
     for (int i = 0; i < nsl; i++) {
       const Slice& s = *list[i];
       const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
-      qd[i] = QuantDesc{s.data, out_base + s.comp_off, s.n, s.bucket, flags};
+      void* fb = fb_base ? fb_base + s.fb_off * elem_size(dt) : nullptr;
+      TORCH_CHECK(!fb || s.bucket % 8 == 0,
+                  "cgx: CGX_ERROR_FEEDBACK requires bucket_size % 8 == 0");
+      qd[i] = QuantDesc{s.data, out_base + s.comp_off, fb, s.n, s.bucket,
+                        flags};
       if (s.skip_incomplete) {
         cum[i + 1] = cum[i] + s.n / s.bucket;
         any_residual |= (s.n % s.bucket) != 0;
@@ -414,7 +432,7 @@ Engine::ChunkPlan Engine::plan(const std::vector<LayerView>& views,
       const int64_t hi = std::min(pos + v.numel, end);
       if (hi > lo) {
         pl.rs[r].push_back(Slice{v.data + (lo - pos) * es, hi - lo, v.bits,
-                                 v.bucket_size, coff, skip_incomplete});
+                                 v.bucket_size, coff, skip_incomplete, lo});
         coff += buffer_size(hi - lo, dt, v.bits, v.bucket_size,
                             skip_incomplete);
       }
@@ -526,7 +544,15 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
         all.push_back(t);
       }
     }
-    run_quantize(all, send1, dt, qs, cfg.stochastic);
+    char* fb1 = nullptr;
+    if (cfg.error_feedback) {
+      at::Tensor& t =
+          feedback_buf(views[0].data, pl.n, /*phase=*/1,
+                       dt == DType::F32 ? at::kFloat
+                       : dt == DType::F16 ? at::kHalf : at::kBFloat16);
+      fb1 = static_cast<char*>(t.data_ptr());
+    }
+    run_quantize(all, send1, dt, qs, cfg.stochastic, fb1);
   }
   timer_mark(1, qs);
 
@@ -552,7 +578,17 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
                 deq_stream_);
     // self-quantize the reduced chunk; the same bytes go to every peer and
     // through my own decode so all ranks end bit-identical
-    run_quantize(rs[rank_], send2, dt, deq_stream_, cfg.stochastic);
+    char* fb2 = nullptr;
+    if (cfg.error_feedback && pl.szs[rank_] > 0) {
+      at::Tensor& t =
+          feedback_buf(views[0].data, pl.szs[rank_], /*phase=*/2,
+                       dt == DType::F32 ? at::kFloat
+                       : dt == DType::F16 ? at::kHalf : at::kBFloat16);
+      // fb_off is in chunk space; rebase to this rank's chunk start
+      fb2 = static_cast<char*>(t.data_ptr()) -
+            pl.offs[rank_] * elem_size(dt);
+    }
+    run_quantize(rs[rank_], send2, dt, deq_stream_, cfg.stochastic, fb2);
   }
   timer_mark(3, deq_stream_);
 

@@ -15,7 +15,9 @@
 #include <rccl/rccl.h>
 
 #include <cstdint>
+#include <map>
 #include <mutex>
+#include <tuple>
 #include <unordered_map>
 #include <vector>
 
@@ -96,6 +98,7 @@ struct EngineConfig {
   bool skip_incomplete = false;  // CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS
   bool dummy = false;  // CGX_DEBUG_DUMMY_COMPRESSION: force uncompressed
   bool intra_compress = true;  // CGX_INTRA_COMPRESS
+  bool error_feedback = false;  // CGX_ERROR_FEEDBACK (needs bucket%8==0)
   static EngineConfig from_env();  // re-read every bucket like the reference
 };
 
@@ -173,6 +176,7 @@ class Engine {
     int bucket;
     int64_t comp_off;  // byte offset of this slice in the chunk's comp stream
     bool skip_incomplete = false;
+    int64_t fb_off = 0;  // element offset in the chunk's error-feedback buf
   };
 
   // Double-buffered staging so chunk c+1's quantize (on qs) can start while
@@ -203,8 +207,14 @@ class Engine {
   hipEvent_t next_ev();
 
   // Launch one quantize "job list", grouping slices by (bits, bucket%8==0).
+  // fb_base (optional): error-feedback buffer in chunk element space.
   void run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
-                    DType dt, hipStream_t stream, bool stochastic);
+                    DType dt, hipStream_t stream, bool stochastic,
+                    char* fb_base = nullptr);
+  // per-(chunk, phase) persistent error-feedback buffers
+  at::Tensor& feedback_buf(const void* key_ptr, int64_t numel, int phase,
+                           at::ScalarType st);
+  std::map<std::tuple<const void*, int64_t, int>, at::Tensor> fb_bufs_;
   void run_dequant(const std::vector<Slice>& slices, const uint8_t* in_base,
                    int64_t src_stride, int nsrc, bool add, DType dt,
                    hipStream_t stream);

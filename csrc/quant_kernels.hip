@@ -349,7 +349,8 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     float lmin = INFINITY, lmax = -INFINITY;
     // fast path: full bucket, whole groups per lane (wave-uniform branch)
     const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
-    if (full && ngroups <= MAXG * kWave) {
+    T* const fbp = d.fb ? reinterpret_cast<T*>(d.fb) + bstart : nullptr;
+    if (full && ngroups <= MAXG * kWave && !fbp) {
       if constexpr (sizeof(T) == 2) {
         using PK = Pk2<T>;
         typename PK::P pmin = bitcast<typename PK::P>(PK::kInf);
@@ -383,6 +384,21 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
           }
         }
       }
+    } else if (full && ngroups <= MAXG * kWave && fbp) {
+      // error-feedback: stash the T-rounded x+fb and reduce over it
+      for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
+        uint32_t xr[8], fr[8];
+        load8<T>(in + g * 8, true, xr);
+        load8<T>(fbp + g * 8, al16, fr);
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          const uint32_t xe = f2raw<T>(raw2f<T>(xr[j]) + raw2f<T>(fr[j]));
+          stash[gi][j] = xe;
+          const float f = raw2f<T>(xe);
+          lmin = fminf(lmin, f);
+          lmax = fmaxf(lmax, f);
+        }
+      }
     } else {
       for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
         uint32_t r[8];
@@ -393,6 +409,11 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
           const R* q = reinterpret_cast<const R*>(in) + g * 8;
           for (int j = 0; j < m; j++) r[j] = q[j];
           for (int j = m; j < 8; j++) r[j] = 0;
+        }
+        if (fbp) {
+          const R* q = reinterpret_cast<const R*>(fbp) + g * 8;
+          for (int j = 0; j < m; j++)
+            r[j] = f2raw<T>(raw2f<T>(r[j]) + raw2f<T>(q[j]));
         }
         if (gi < MAXG) {
 #pragma unroll
@@ -422,7 +443,7 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
       const int64_t gbase = bstart >> 3;
       const bool live = unitf >= kEps;
       const float rinv = 1.0f / unitf;  // hoisted: fp32 div is 1/4 VALU rate
-      if (full && ngroups <= MAXG * kWave && live) {
+      if (full && ngroups <= MAXG * kWave && live && !fbp) {
         // 8 levels fit a uint32 when BITS <= 4: halves the shift/or cost
         using acc_t =
             typename std::conditional<(BITS <= 4), uint32_t, uint64_t>::type;
@@ -453,13 +474,19 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
         if (gi < MAXG) {
 #pragma unroll
           for (int j = 0; j < 8; j++) r[j] = stash[gi][j];
-        } else if (m == 8) {
+        } else if (m == 8 && !fbp) {
           load8<T>(in + g * 8, al16, r);
         } else {
           const R* q = reinterpret_cast<const R*>(in) + g * 8;
           for (int j = 0; j < m; j++) r[j] = q[j];
+          if (fbp) {
+            const R* f = reinterpret_cast<const R*>(fbp) + g * 8;
+            for (int j = 0; j < m; j++)
+              r[j] = f2raw<T>(raw2f<T>(r[j]) + raw2f<T>(f[j]));
+          }
         }
         uint64_t value = 0;
+        uint32_t lv[8] = {};
         if (live) {
           const uint64_t pr =
               stochastic ? rand_pack(seed, (static_cast<uint64_t>(lo) << 44) |
@@ -470,8 +497,20 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
             const float dd = (raw2f<T>(r[j]) - minf) * rinv + rnd;
             const uint32_t level =
                 static_cast<uint32_t>(fminf(floorf(dd), divisor));
+            lv[j] = level;
             value |= static_cast<uint64_t>(level & ((1u << BITS) - 1))
                      << (j * BITS);
+          }
+        }
+        if (fbp) {
+          // residual: (x+fb) - decode(level), rounded to T per the spec
+          R* f = reinterpret_cast<R*>(fbp) + g * 8;
+          for (int j = 0; j < m; j++) {
+            const uint32_t prod =
+                f2raw<T>(unitf * static_cast<float>(live ? lv[j] : 0));
+            const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+            f[j] = static_cast<R>(
+                f2raw<T>(raw2f<T>(r[j]) - raw2f<T>(dec)));
           }
         }
         const int64_t gb = (gbase + g) * BITS;
@@ -788,7 +827,18 @@ __global__ __launch_bounds__(kThreads) void k_residual_q(
     const R* src = reinterpret_cast<const R*>(d.in) + nq;
     R* dst = reinterpret_cast<R*>(
         d.out + 2 * sizeof(R) * nb + ((nq * bits + 7) / 8 + 7) / 8 * 8);
-    for (int64_t i = t0; i < r; i += stride) dst[i] = src[i];
+    if (d.fb) {
+      // error feedback: the raw residual tail carries x+fb exactly, so the
+      // new residual is zero
+      R* f = reinterpret_cast<R*>(d.fb) + nq;
+      for (int64_t i = t0; i < r; i += stride) {
+        dst[i] = static_cast<R>(
+            f2raw<T>(raw2f<T>(src[i]) + raw2f<T>(f[i])));
+        f[i] = 0;
+      }
+    } else {
+      for (int64_t i = t0; i < r; i += stride) dst[i] = src[i];
+    }
   }
 }
 

@@ -105,3 +105,22 @@ def test_skip_incomplete_roundtrip(dtype, n, bucket):
         tol = unit * 1.05 + 1e-6 + (0 if dtype == torch.float32 else
                                     xb.abs().max() * 2 ** -7)
         assert (err <= tol).all()
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16])
+def test_error_feedback_time_average_converges(dtype):
+    torch.manual_seed(3)
+    x = torch.randn(2048).to(dtype)
+    fb = torch.zeros_like(x)
+    acc = torch.zeros(2048)
+    reps = 20
+    for k in range(1, reps + 1):
+        comp = golden.quantize_ef(x, fb, 2, 512)
+        acc += golden.dequantize(comp, x.numel(), dtype, 2, 512).float()
+    ef_err = ((acc / reps) - x.float()).abs().max().item()
+    one_shot = (golden.dequantize(golden.quantize(x, 2, 512), x.numel(),
+                                  dtype, 2, 512).float()
+                - x.float()).abs().max().item()
+    # EF makes the *time-averaged* transmission unbiased: much closer than a
+    # single deterministic quantization
+    assert ef_err < one_shot * 0.35, (ef_err, one_shot)

@@ -195,3 +195,40 @@ def test_skip_incomplete_dequant_add():
     out = base.to(_dev()).clone()
     _C.dequantize(comp.to(_dev()), out, bits, bucket, True, True)
     assert torch.equal(out.cpu(), expected)
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("bits", [2, 4, 8])
+@pytest.mark.parametrize("n,bucket", [(4096, 512), (1000, 512), (131, 64),
+                                      (4096, 1024)])
+def test_error_feedback_matches_golden(dtype, bits, n, bucket):
+    from torch_cgx_amd import _C
+    torch.manual_seed(n + bits)
+    x = torch.randn(n).to(dtype)
+    fb_gold = (torch.randn(n) * 0.1).to(dtype)
+    fb_gpu = fb_gold.clone().to(_dev())
+    comp_gold = golden.quantize_ef(x.clone(), fb_gold, bits, bucket)
+    comp_gpu = _C.quantize(x.to(_dev()), bits, bucket, False, 0, False,
+                           fb_gpu).cpu()
+    assert torch.equal(comp_gpu, comp_gold)
+    assert torch.equal(fb_gpu.cpu(), fb_gold)
+
+
+def test_error_feedback_repeated_use():
+    from torch_cgx_amd import _C
+    torch.manual_seed(11)
+    n, bits, bucket = 1 << 16, 2, 512
+    x = torch.randn(n, device=_dev())
+    fb = torch.zeros_like(x)
+    acc = torch.zeros_like(x)
+    reps = 16
+    out = torch.empty_like(x)
+    for _ in range(reps):
+        comp = _C.quantize(x, bits, bucket, False, 0, False, fb)
+        _C.dequantize(comp, out, bits, bucket, False)
+        acc += out
+    ef_err = ((acc / reps) - x).abs().max().item()
+    one = _C.quantize(x, bits, bucket, False, 0)
+    _C.dequantize(one, out, bits, bucket, False)
+    one_err = (out - x).abs().max().item()
+    assert ef_err < one_err * 0.35, (ef_err, one_err)

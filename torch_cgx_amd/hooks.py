@@ -44,6 +44,10 @@ class CGXState:
                 "bits", self.quantization_bits)
             self.quantization_bucket_size = compression_params.get(
                 "bucket_size", self.quantization_bucket_size)
+            if compression_params.get("error_feedback"):
+                # engine config is env-driven (re-read every bucket), so
+                # every rank picks this up consistently
+                os.environ["CGX_ERROR_FEEDBACK"] = "1"
 
     def should_compress_(self, tensor: torch.Tensor) -> bool:
         if tensor.dim() <= 1 or tensor.numel() < self.layer_min_size:

@@ -164,6 +164,22 @@ def _from_raw_bytes(b: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     return torch.from_numpy(arr.view(np.int16).copy()).view(dtype)
 
 
+def quantize_ef(x: torch.Tensor, feedback: torch.Tensor, bits: int,
+                bucket_size: int, rand=0.5) -> torch.Tensor:
+    """Error-feedback quantize: encodes x + feedback and updates feedback in
+    place with the new residual (all in fp32, residual rounded to x.dtype).
+
+    The reference shipped EF kernel plumbing but never enabled it
+    (cuda_compression_operations.cu:713-725 passes EF=false); this is the
+    working version.  Wire format is unchanged."""
+    assert feedback.shape == x.shape and feedback.dtype == x.dtype
+    xe = (x.float() + feedback.float()).to(x.dtype)
+    comp = quantize(xe, bits, bucket_size, rand)
+    dec = dequantize(comp, xe.numel(), x.dtype, bits, bucket_size)
+    feedback.copy_((xe.float() - dec.float()).to(x.dtype))
+    return comp
+
+
 def quantize(x: torch.Tensor, bits: int, bucket_size: int,
              rand=0.5, skip_incomplete: bool = False) -> torch.Tensor:
     """Compress x -> uint8 buffer of exactly buffer_size(...) bytes.
