@@ -105,3 +105,26 @@ def _registry_snapshot_worker(rank, ws):
 
 def test_registry_snapshot_after_hook():
     run_dist(_registry_snapshot_worker, 2)
+
+
+def _adaptive_worker(rank, ws):
+    import torch_cgx_amd
+    from torch_cgx_amd import _C
+    from torch_cgx_amd.adaptive import AdaptiveCGXState, adaptive_cgx_hook
+    _C.clear_registry()
+    state = AdaptiveCGXState(None, schedule=[(0, 8), (4, 2)],
+                             layer_min_size=32, bucket_size=128)
+    _train(Net(), 8, rank, adaptive_cgx_hook, state)
+    snap = _C.registry_snapshot()
+    assert snap
+    saw2 = False
+    for (_i, _n, cfgs) in snap:
+        for (bits, _bs) in cfgs:
+            assert bits in (2, 32)  # schedule reached 2-bit; fp32 pins stay
+            saw2 |= bits == 2
+    assert saw2
+    _C.clear_registry()
+
+
+def test_adaptive_schedule():
+    run_dist(_adaptive_worker, 2)

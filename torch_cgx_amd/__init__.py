@@ -68,8 +68,10 @@ set_quantization_bits = _C.set_quantization_bits
 set_quantization_bucket_size = _C.set_quantization_bucket_size
 
 from .hooks import CGXState, cgx_hook  # noqa: E402
+from .adaptive import AdaptiveCGXState, adaptive_cgx_hook  # noqa: E402
 
 __all__ = [
-    "CGXState", "cgx_hook", "register_layer",
-    "set_quantization_bits", "set_quantization_bucket_size",
+    "CGXState", "cgx_hook", "AdaptiveCGXState", "adaptive_cgx_hook",
+    "register_layer", "set_quantization_bits",
+    "set_quantization_bucket_size",
 ]
