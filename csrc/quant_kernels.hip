@@ -577,6 +577,23 @@ __global__ __launch_bounds__(kThreads) void k_pack_generic(
   }
 }
 
+// Load a (unit, min) meta pair with ONE aligned load (pairs are naturally
+// 2*sizeof(R)-aligned: the slice base is 8-byte aligned).
+template <typename R>
+__device__ __forceinline__ void load_meta_pair(const R* meta, int64_t bk,
+                                               uint32_t& unit_raw,
+                                               uint32_t& min_raw) {
+  if constexpr (sizeof(R) == 4) {
+    const uint64_t p = *reinterpret_cast<const uint64_t*>(meta + 2 * bk);
+    unit_raw = static_cast<uint32_t>(p);
+    min_raw = static_cast<uint32_t>(p >> 32);
+  } else {
+    const uint32_t p = *reinterpret_cast<const uint32_t*>(meta + 2 * bk);
+    unit_raw = p & 0xFFFF;
+    min_raw = p >> 16;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Dequantize(+multi-source accumulate): grid-stride threads over packs.
 // Sums d.nsrc compressed streams in T precision in stream order (matching the
@@ -692,15 +709,20 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
           } else {
             value = load_bytes(src + g * BITS, BITS) >> (h * 4 * BITS);
           }
+          uint32_t u0 = 0, m0 = 0;
+          if (oneb) load_meta_pair<R>(meta, bk0, u0, m0);
 #pragma unroll
           for (int j = 0; j < 4; j++) {
-            const int64_t bk = oneb ? bk0 : (w * 4 + j) / d.bucket;
+            uint32_t ur = u0, mr = m0;
+            if (!oneb) {
+              const int64_t bk = (w * 4 + j) / d.bucket;
+              load_meta_pair<R>(meta, bk, ur, mr);
+            }
             const uint32_t lvl = static_cast<uint32_t>(
                 (value >> (j * BITS)) & ((1u << BITS) - 1));
-            const float unitf = raw2f<T>(meta[2 * bk]);
-            const float minf = raw2f<T>(meta[2 * bk + 1]);
-            const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
-            const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+            const uint32_t prod =
+                f2raw<T>(raw2f<T>(ur) * static_cast<float>(lvl));
+            const uint32_t dec = f2raw<T>(raw2f<T>(mr) + raw2f<T>(prod));
             if (!have && sidx == 0) {
               v[j] = dec;
             } else {
@@ -756,15 +778,19 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
         const uint8_t* src = in0 + sidx * d.src_stride;
         const R* meta = reinterpret_cast<const R*>(src - meta_bytes);
         const uint64_t value = load_bytes(src + g * BITS, BITS);
+        uint32_t u0 = 0, m0 = 0;
+        if (oneb) load_meta_pair<R>(meta, bk0, u0, m0);
 #pragma unroll
         for (int j = 0; j < 8; j++) {
-          const int64_t bk = oneb ? bk0 : (g * 8 + j) / d.bucket;
+          uint32_t ur = u0, mr = m0;
+          if (!oneb) {
+            const int64_t bk = (g * 8 + j) / d.bucket;
+            load_meta_pair<R>(meta, bk, ur, mr);
+          }
           const uint32_t lvl = static_cast<uint32_t>((value >> (j * BITS)) &
                                                      ((1u << BITS) - 1));
-          const float unitf = raw2f<T>(meta[2 * bk]);
-          const float minf = raw2f<T>(meta[2 * bk + 1]);
-          const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
-          const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+          const uint32_t prod = f2raw<T>(raw2f<T>(ur) * static_cast<float>(lvl));
+          const uint32_t dec = f2raw<T>(raw2f<T>(mr) + raw2f<T>(prod));
           if (!have && sidx == 0) {
             v[j] = dec;
           } else {
