@@ -1,15 +1,17 @@
 // CDNA4 (gfx950) max-min quantization kernels for the cgx backend.
 //
 // Design (MI355X-native, not a port):
-//  * wavefront = 64 lanes; one WAVE per quantization bucket — for the default
-//    bucket sizes (512/1024) each lane owns exactly 1-2 packs of 8 values,
-//    so the min/max reduction is a pure in-register __shfl_xor butterfly
-//    (no LDS) and the values stay in registers between the meta and encode
-//    phases (single HBM read).
+//  * wavefront = 64 lanes; one WAVE per quantization bucket — for buckets up
+//    to 2048 every lane's packs stay in registers between the meta and
+//    encode phases (single HBM read), the min/max reduction is a DPP row-op
+//    sequence (register path; no LDS, no ds_bpermute), and 16-bit dtypes
+//    reduce with packed v_pk_min/max.
 //  * memory-bound workload: 16-byte vectorized loads/stores wherever the
-//    slice base is 16B-aligned; grid-stride loops sized ≫ 256 CUs.
-//  * stochastic rounding via a stateless splitmix64 hash (no RNG state
-//    buffers, fully deterministic given the per-launch seed).
+//    slice base is 16B-aligned; decode uses 4 elems/thread for fp32 so every
+//    store is one coalesced float4; grid-stride loops sized >> 256 CUs.
+//  * stochastic rounding via a stateless splitmix64 hash per pack (no RNG
+//    state buffers, deterministic given the per-launch seed); optional
+//    error-feedback residual update fused into the encode.
 //
 // Wire-format parity with the reference implementation is defined by
 // torch_cgx_amd/ops/golden.py (see reference
