@@ -216,7 +216,8 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
         ncclCommInitRank(&cross_comm_, topo_.n_nodes, cid, topo_.node_id));
     intra_engine_ = std::make_unique<Engine>(topo_.local_rank,
                                              topo_.local_size);
-    cross_engine_ = std::make_unique<Engine>(topo_.node_id, topo_.n_nodes);
+    cross_engine_ = std::make_unique<Engine>(topo_.node_id, topo_.n_nodes,
+                                             /*is_cross=*/true);
     hierarchical_ = true;
   }
   if (env_flag("CGX_VERBOSE") && rank_ == 0) {
@@ -316,7 +317,8 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
         // registry is consulted exactly ONCE per bucket so the cursor stays
         // in lockstep on every rank (leaders run two engine passes).
         cgx::Registry::BucketInfo info;
-        const bool matched = cgx::Registry::get().next(t.numel(), &info);
+        const bool matched =
+            cgx::Registry::get().next(t.numel(), t.data_ptr(), &info);
         fin = intra_engine_->allreduce(t, intra_comm_, stream_->stream(),
                                        &info, matched);
         if (topo_.local_rank == 0)
@@ -556,6 +558,14 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::reduce_scatter(
   auto op = opts.reduceOp;
   return collective(outputs, out.device(), c10d::OpType::REDUCE_SCATTER,
                     [this, flat, out, op](hipStream_t s) {
+                      // flat is a temporary destroyed when this call returns;
+                      // keep its storage out of the allocator's reuse pool
+                      // until the async collective on the side stream is done
+                      c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
+                          recordStreamMasqueradingAsCUDA(
+                              flat.storage().data_ptr(),
+                              c10::hip::getStreamFromExternalMasqueradingAsCUDA(
+                                  s, flat.device().index()));
                       CGX_NCCL_CHECK(ncclReduceScatter(
                           flat.data_ptr(), out.data_ptr(), out.numel(),
                           nccl_dtype(out), to_nccl_op(op), comm_, s));

@@ -183,6 +183,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_quantization_bits", &cgx::py_set_bits);
   m.def("set_quantization_bucket_size", &cgx::py_set_bucket);
   m.def("clear_registry", [] { cgx::Registry::get().clear(); });
+  m.def("registry_match",
+        [](int64_t numel, int64_t key) {
+          // test seam for Registry::next: key is an opaque bucket-storage
+          // identity (the engine passes the flat tensor's data_ptr)
+          cgx::Registry::BucketInfo info;
+          const bool ok = cgx::Registry::get().next(
+              numel, reinterpret_cast<const void*>(key), &info);
+          return py::make_tuple(ok, ok ? info.idx : -1);
+        },
+        py::arg("numel"), py::arg("key") = 0);
   m.def("registry_snapshot", [] {
     // [(bucket_idx, [numels], [(bits, bucket_size)])] in registration order
     py::list out;
@@ -197,6 +207,27 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return out;
   });
 
+  m.def("parse_engine_config", [] {
+    // test seam: the env-derived engine configuration, as each Engine
+    // re-reads it per bucket (inner_ring/cross_ring select the reduction
+    // algorithm for the intra-node vs cross-node engine independently)
+    auto c = cgx::EngineConfig::from_env();
+    py::dict d;
+    d["fusion_bytes"] = c.fusion_bytes;
+    d["min_elems"] = c.min_elems;
+    d["default_bits"] = c.default_bits;
+    d["default_bucket"] = c.default_bucket;
+    d["stochastic"] = c.stochastic;
+    d["inner_ring"] = c.ring;
+    d["cross_ring"] = c.cross_ring;
+    d["debug_a2a"] = c.debug_a2a;
+    d["fake_ratio"] = c.fake_ratio;
+    d["skip_incomplete"] = c.skip_incomplete;
+    d["dummy"] = c.dummy;
+    d["intra_compress"] = c.intra_compress;
+    d["error_feedback"] = c.error_feedback;
+    return d;
+  });
   m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
         py::arg("bucket_size"), py::arg("stochastic") = false,
         py::arg("seed") = 0, py::arg("skip_incomplete") = false,

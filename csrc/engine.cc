@@ -67,6 +67,7 @@ void Registry::clear() {
   std::lock_guard<std::mutex> g(mu_);
   order_.clear();
   by_idx_.clear();
+  bound_.clear();
   cursor_ = 0;
 }
 
@@ -80,9 +81,23 @@ std::vector<Registry::BucketInfo> Registry::snapshot() {
   return order_;
 }
 
-bool Registry::next(int64_t numel, BucketInfo* out) {
+bool Registry::next(int64_t numel, const void* key, BucketInfo* out) {
   std::lock_guard<std::mutex> g(mu_);
   if (order_.empty()) return false;
+  // pointer identity first: a previously seen flat-bucket storage resolves
+  // to the bucket it matched before, regardless of cursor position
+  if (key) {
+    auto it = bound_.find(key);
+    if (it != bound_.end()) {
+      const size_t i = it->second;
+      if (i < order_.size() && order_[i].total == numel) {
+        *out = order_[i];
+        cursor_ = i + 1;
+        return true;
+      }
+      bound_.erase(it);  // bucket rebuilt or storage reused: unlearn
+    }
+  }
   const size_t n = order_.size();
   const size_t start = cursor_ % n;
   // prefer the cursor position; otherwise any bucket whose total matches
@@ -91,6 +106,7 @@ bool Registry::next(int64_t numel, BucketInfo* out) {
     if (order_[i].total == numel) {
       *out = order_[i];
       cursor_ = i + 1;
+      if (key) bound_[key] = i;
       return true;
     }
   }
@@ -119,14 +135,24 @@ EngineConfig EngineConfig::from_env() {
   c.dummy = env_int("CGX_DEBUG_DUMMY_COMPRESSION", 0) != 0;
   c.intra_compress = env_int("CGX_INTRA_COMPRESS", 1) != 0;
   c.error_feedback = env_int("CGX_ERROR_FEEDBACK", 0) != 0;
+  c.debug_a2a = env_int("CGX_DEBUG_ALL_TO_ALL_REDUCTION", 0) != 0;
   const char* fr = std::getenv("CGX_COMPRESSION_FAKE_RATIO");
   if (fr && *fr) c.fake_ratio = std::atof(fr);
   if (!(c.fake_ratio > 0.0 && c.fake_ratio <= 1.0)) c.fake_ratio = 1.0;
-  const char* red = std::getenv("CGX_INNER_REDUCTION_TYPE");
-  if (!red || !*red) red = std::getenv("CGX_REDUCTION_TYPE");
-  c.ring = red && (std::strcmp(red, "Ring") == 0 ||
-                   std::strcmp(red, "ring") == 0 ||
-                   std::strcmp(red, "RING") == 0);
+  // Reduction algorithm selection, independently for the intra-node and
+  // cross-node engines (reference mpi_allreduce_operations.cc:90-115:
+  // intra default SRA, cross default Ring).  The legacy CGX_REDUCTION_TYPE
+  // alias drives whichever specific variable is unset.
+  auto is_ring = [](const char* v, bool dflt) {
+    if (!v || !*v) return dflt;
+    return std::strcmp(v, "Ring") == 0 || std::strcmp(v, "ring") == 0 ||
+           std::strcmp(v, "RING") == 0;
+  };
+  const char* legacy = std::getenv("CGX_REDUCTION_TYPE");
+  const char* inner = std::getenv("CGX_INNER_REDUCTION_TYPE");
+  const char* cross = std::getenv("CGX_CROSS_REDUCTION_TYPE");
+  c.ring = is_ring(inner && *inner ? inner : legacy, false);
+  c.cross_ring = is_ring(cross && *cross ? cross : legacy, true);
   return c;
 }
 
@@ -200,7 +226,8 @@ ncclDataType_t nccl_dtype(const at::Tensor& t) {
 // ---------------------------------------------------------------------------
 // Engine
 // ---------------------------------------------------------------------------
-Engine::Engine(int rank, int size) : rank_(rank), size_(size) {
+Engine::Engine(int rank, int size, bool is_cross)
+    : rank_(rank), size_(size), is_cross_(is_cross) {
   seed_ = 0x9E3779B97F4A7C15ull ^ (0xD1B54A32D192ED03ull * (uint64_t)(rank + 1));
   // constructed under the backend's device guard (lazyInit)
   int least = 0, greatest = 0;
@@ -319,21 +346,35 @@ uint8_t* Engine::slot_bytes(StagingSlot& slot, int64_t bytes) {
   return slot.buf.data_ptr<uint8_t>();
 }
 
-at::Tensor& Engine::feedback_buf(const void* key_ptr, int64_t numel,
-                                 int phase, at::ScalarType st) {
-  auto key = std::make_tuple(key_ptr, numel, phase);
-  auto it = fb_bufs_.find(key);
+at::Tensor& Engine::feedback_buf(int64_t key, int phase, int64_t numel,
+                                 at::ScalarType st) {
+  // bounded: residuals are a convergence aid; resetting them once (on a cap
+  // overflow that only a pathological caller can trigger) is benign,
+  // unbounded growth is not
+  constexpr size_t kMaxEntries = 4096;
+  if (fb_bufs_.size() >= kMaxEntries) {
+    TORCH_WARN_ONCE("cgx: error-feedback store exceeded ", kMaxEntries,
+                    " entries; resetting residuals");
+    fb_bufs_.clear();
+  }
+  auto k = std::make_pair(key, phase);
+  auto it = fb_bufs_.find(k);
+  if (it != fb_bufs_.end() &&
+      (it->second.numel() < numel || it->second.scalar_type() != st)) {
+    fb_bufs_.erase(it);  // chunk grew or dtype changed: restart residual
+    it = fb_bufs_.end();
+  }
   if (it == fb_bufs_.end()) {
     auto t = at::zeros({numel}, at::TensorOptions().dtype(st).device(
                                     at::kCUDA));
-    it = fb_bufs_.emplace(key, std::move(t)).first;
+    it = fb_bufs_.emplace(k, std::move(t)).first;
   }
   return it->second;
 }
 
 void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
                           DType dt, hipStream_t stream, bool stochastic,
-                          char* fb_base) {
+                          char* fb_base, int64_t fb_rebase) {
   // group by (bits, bucket%8==0)
   std::map<std::pair<int, bool>, std::vector<const Slice*>> groups;
   for (const auto& s : slices) {
@@ -352,7 +393,10 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
     for (int i = 0; i < nsl; i++) {
       const Slice& s = *list[i];
       const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
-      void* fb = fb_base ? fb_base + s.fb_off * elem_size(dt) : nullptr;
+      void* fb =
+          fb_base ? fb_base + (s.fb_off - fb_rebase) * elem_size(dt) : nullptr;
+      TORCH_CHECK(!fb_base || s.fb_off >= fb_rebase,
+                  "cgx: error-feedback slice offset below buffer base");
       TORCH_CHECK(!fb || s.bucket % 8 == 0,
                   "cgx: CGX_ERROR_FEEDBACK requires bucket_size % 8 == 0");
       qd[i] = QuantDesc{s.data, out_base + s.comp_off, fb, s.n, s.bucket,
@@ -496,7 +540,7 @@ void Engine::timer_finish() {
 
 void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
                        ncclComm_t comm, hipStream_t qs,
-                       const EngineConfig& cfg) {
+                       const EngineConfig& cfg, int64_t fb_key) {
   const int ws = size_;
   ChunkPlan pl = plan(views, dt, cfg.skip_incomplete);
   if (pl.n == 0) return;
@@ -547,7 +591,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
     char* fb1 = nullptr;
     if (cfg.error_feedback) {
       at::Tensor& t =
-          feedback_buf(views[0].data, pl.n, /*phase=*/1,
+          feedback_buf(fb_key, /*phase=*/1, pl.n,
                        dt == DType::F32 ? at::kFloat
                        : dt == DType::F16 ? at::kHalf : at::kBFloat16);
       fb1 = static_cast<char*>(t.data_ptr());
@@ -579,16 +623,20 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
     // self-quantize the reduced chunk; the same bytes go to every peer and
     // through my own decode so all ranks end bit-identical
     char* fb2 = nullptr;
+    int64_t fb2_rebase = 0;
     if (cfg.error_feedback && pl.szs[rank_] > 0) {
       at::Tensor& t =
-          feedback_buf(views[0].data, pl.szs[rank_], /*phase=*/2,
+          feedback_buf(fb_key, /*phase=*/2, pl.szs[rank_],
                        dt == DType::F32 ? at::kFloat
                        : dt == DType::F16 ? at::kHalf : at::kBFloat16);
-      // fb_off is in chunk space; rebase to this rank's chunk start
-      fb2 = static_cast<char*>(t.data_ptr()) -
-            pl.offs[rank_] * elem_size(dt);
+      // slice fb_off is in chunk element space; the phase-2 buffer covers
+      // only this rank's partition, so rebase by its start (host-side per
+      // slice — no out-of-range pointer arithmetic)
+      fb2 = static_cast<char*>(t.data_ptr());
+      fb2_rebase = pl.offs[rank_];
     }
-    run_quantize(rs[rank_], send2, dt, deq_stream_, cfg.stochastic, fb2);
+    run_quantize(rs[rank_], send2, dt, deq_stream_, cfg.stochastic, fb2,
+                 fb2_rebase);
   }
   timer_mark(3, deq_stream_);
 
@@ -636,6 +684,10 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
   // ring.cc:139-226): ws-1 reduce-scatter steps with per-hop requantize of
   // the running partial sum, then ws-1 allgather steps FORWARDING the
   // once-quantized reduced segments, final batch decompress.
+  TORCH_CHECK(!cfg.error_feedback,
+              "cgx: CGX_ERROR_FEEDBACK is not supported with Ring reduction "
+              "(the ring requantizes running partial sums, which have no "
+              "stable per-step residual); use SRA or disable error feedback");
   const int ws = size_;
   ChunkPlan pl = plan(views, dt, cfg.skip_incomplete);
   if (pl.n == 0) return;
@@ -705,6 +757,54 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
   run_dequant(all, segs, 0, 1, /*add=*/false, dt, stream);
 }
 
+void Engine::a2a_chunk(const std::vector<LayerView>& views, DType dt,
+                       ncclComm_t comm, hipStream_t qs,
+                       const EngineConfig& cfg) {
+  // Debug brute-force reduction (reference AllReduceAlltoAllCompressed,
+  // scatter_reduce_allgather.cc:269-306): every rank quantizes its ENTIRE
+  // chunk once and sends it to every peer; all ranks then decode the same
+  // ws compressed streams (self first, overwrite; peers accumulated) so the
+  // result is bit-identical across ranks.  ws x the wire bytes of SRA, but
+  // no partitioning/offset machinery in the fault surface.
+  TORCH_CHECK(!cfg.error_feedback,
+              "cgx: CGX_DEBUG_ALL_TO_ALL_REDUCTION does not support "
+              "CGX_ERROR_FEEDBACK");
+  const int ws = size_;
+  std::vector<Slice> sl;
+  int64_t coff = 0, pos = 0;
+  for (const auto& v : views) {
+    if (v.numel <= 0) continue;
+    sl.push_back(Slice{v.data, v.numel, v.bits, v.bucket_size, coff,
+                       cfg.skip_incomplete, pos});
+    coff +=
+        buffer_size(v.numel, dt, v.bits, v.bucket_size, cfg.skip_incomplete);
+    pos += v.numel;
+  }
+  if (coff == 0) return;
+  const int64_t C = coff;
+  uint8_t* base = staging(C * ws, qs);
+  uint8_t* send = base;          // own compressed stream
+  uint8_t* recv = base + C;      // ws-1 peer streams, stride C
+  run_quantize(sl, send, dt, qs, cfg.stochastic);
+  chain(qs, comm_stream_);
+  CGX_NCCL_CHECK(ncclGroupStart());
+  for (int p = 0; p < ws; p++) {
+    if (p == rank_) continue;
+    const int s = p < rank_ ? p : p - 1;
+    CGX_NCCL_CHECK(ncclSend(send, C, ncclUint8, p, comm, comm_stream_));
+    CGX_NCCL_CHECK(ncclRecv(recv + (int64_t)s * C, C, ncclUint8, p, comm,
+                            comm_stream_));
+  }
+  CGX_NCCL_CHECK(ncclGroupEnd());
+  chain(comm_stream_, deq_stream_);
+  run_dequant(sl, send, 0, 1, /*add=*/false, dt, deq_stream_);
+  if (ws > 1)
+    run_dequant(sl, recv, C, ws - 1, /*add=*/true, dt, deq_stream_);
+  // the shared staging() buffer is reused by the next chunk's quantize on
+  // qs: fence it behind this chunk's decode
+  chain(deq_stream_, qs);
+}
+
 hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
                               hipStream_t qs) {
   const EngineConfig cfg = EngineConfig::from_env();
@@ -744,7 +844,8 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
                               bool forced_match) {
   if (size_ <= 1) return qs;
   TORCH_CHECK(bucket.is_contiguous(), "cgx: bucket must be contiguous");
-  const EngineConfig cfg = EngineConfig::from_env();
+  EngineConfig cfg = EngineConfig::from_env();
+  if (is_cross_) cfg.ring = cfg.cross_ring;
   const DType dt = dtype_of(bucket);
   const int es = elem_size(dt);
   char* base = static_cast<char*>(bucket.data_ptr());
@@ -756,7 +857,7 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
     matched = forced_match;
     if (matched) info = *forced;
   } else {
-    matched = Registry::get().next(numel, &info);
+    matched = Registry::get().next(numel, bucket.data_ptr(), &info);
   }
   std::vector<LayerView> views;
   if (matched) {
@@ -805,8 +906,16 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   int64_t cur_n = 0;
   static const bool timings_on = env_int("CGX_TIMINGS", 0) != 0;
   bool any_comp = false;
+  int chunk_seq = 0;
   auto run_chunk = [&](const std::vector<LayerView>& vs_in) {
     any_comp = true;
+    // error-feedback residual identity for this (bucket, chunk): registry
+    // bucket index + chunk ordinal when matched; pointer-derived negative
+    // fallback key otherwise (DDP flat buffers are pointer-stable)
+    const int64_t fb_key =
+        matched ? (((int64_t)info.idx << 20) | chunk_seq)
+                : -(int64_t)(reinterpret_cast<uintptr_t>(vs_in[0].data) >> 3);
+    chunk_seq++;
     timer_begin(qs, timings_on && !cfg.ring);
     // CGX_COMPRESSION_FAKE_RATIO < 1: reduce only a fraction of each chunk
     // (bandwidth experiments; intentionally lossy -- reference
@@ -827,11 +936,18 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
       vsp = &trimmed;
     }
     const std::vector<LayerView>& vs = *vsp;
-    if (cfg.ring && size_ > 2) {
-      ring_chunk(vs, dt, comm, qs, cfg);  // ring is hop-serial: one stream
+    if (cfg.debug_a2a) {
+      a2a_chunk(vs, dt, comm, qs, cfg);
+    } else if (cfg.ring && size_ > 2) {
+      // ring is hop-serial and runs on one stream (qs).  Its NCCL ops must
+      // be ordered behind any uncompressed ncclAllReduce group queued on
+      // comm_stream_ above: unordered concurrent ops on one communicator
+      // are UB in NCCL/RCCL.
+      chain(comm_stream_, qs);
+      ring_chunk(vs, dt, comm, qs, cfg);
       chain(qs, deq_stream_);             // keep completion on deq stream
     } else {
-      sra_chunk(vs, dt, comm, qs, cfg);
+      sra_chunk(vs, dt, comm, qs, cfg, fb_key);
     }
   };
   auto flush = [&]() {

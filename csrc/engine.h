@@ -66,16 +66,21 @@ class Registry {
   // Match the next allreduce call (of `numel` total elements) to a
   // registered bucket, cycling through registration order (the reference's
   // modulo-cursor, extractLayers, mpi_allreduce_operations.cc:257-285 —
-  // hardened: a bucket only matches if its total numel agrees, else we fall
-  // back to whole-tensor default config instead of corrupting the mapping).
+  // hardened twice over: a bucket only matches if its total numel agrees,
+  // and once a flat-bucket storage pointer `key` has matched a bucket, the
+  // pointer is bound to it so later calls resolve by identity even when two
+  // buckets share a total but differ in layer layout (DDP reuses its flat
+  // gradient buffers, so the pointer is stable until a bucket rebuild — at
+  // which point the total check unbinds it and the cursor re-learns).
   // Returns a copy (thread safety) or nullopt-like empty BucketInfo.
-  bool next(int64_t numel, BucketInfo* out);
+  bool next(int64_t numel, const void* key, BucketInfo* out);
   std::vector<BucketInfo> snapshot();
 
  private:
   std::mutex mu_;
   std::vector<BucketInfo> order_;
   std::unordered_map<int, size_t> by_idx_;
+  std::unordered_map<const void*, size_t> bound_;
   size_t cursor_ = 0;
 };
 
@@ -93,7 +98,12 @@ struct EngineConfig {
   int default_bits = 32;
   int default_bucket = 512;
   bool stochastic = true;
-  bool ring = false;  // CGX_INNER_REDUCTION_TYPE=Ring (default SRA)
+  bool ring = false;        // CGX_INNER_REDUCTION_TYPE=Ring (default SRA)
+  bool cross_ring = true;   // CGX_CROSS_REDUCTION_TYPE (default Ring, like
+                            // the reference's cross-node default,
+                            // mpi_allreduce_operations.cc:96-115)
+  bool debug_a2a = false;   // CGX_DEBUG_ALL_TO_ALL_REDUCTION: brute-force
+                            // all-to-all fallback for fault isolation
   double fake_ratio = 1.0;  // CGX_COMPRESSION_FAKE_RATIO (bandwidth expt)
   bool skip_incomplete = false;  // CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS
   bool dummy = false;  // CGX_DEBUG_DUMMY_COMPRESSION: force uncompressed
@@ -128,7 +138,10 @@ class DescRing {
 
 class Engine {
  public:
-  Engine(int rank, int size);
+  // is_cross: this engine drives the cross-node stage of the hierarchical
+  // path and selects its reduction algorithm from CGX_CROSS_REDUCTION_TYPE
+  // (default Ring) instead of CGX_INNER_REDUCTION_TYPE (default SRA).
+  Engine(int rank, int size, bool is_cross = false);
   ~Engine();
 
   // SUM-allreduce of a flat fp32/fp16/bf16 CUDA tensor: registered (or
@@ -196,25 +209,41 @@ class Engine {
   ChunkPlan plan(const std::vector<LayerView>& views, DType dt,
                  bool skip_incomplete);
 
+  // fb_key: stable identity of this (bucket, chunk) for the error-feedback
+  // residual store — (bucket_idx << 20 | chunk ordinal) when the registry
+  // matched, else a pointer-derived fallback key.
   void sra_chunk(const std::vector<LayerView>& views, DType dt,
-                 ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg);
+                 ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg,
+                 int64_t fb_key);
   void ring_chunk(const std::vector<LayerView>& views, DType dt,
                   ncclComm_t comm, hipStream_t stream,
                   const EngineConfig& cfg);
+  // Brute-force debug reduction (CGX_DEBUG_ALL_TO_ALL_REDUCTION parity,
+  // reference scatter_reduce_allgather.cc:269-306): every rank quantizes its
+  // ENTIRE chunk once, sends it to all peers, and every rank decodes the
+  // same ws compressed streams — no partitioning, ws× the wire traffic, but
+  // removes the partition/offset machinery from the fault surface.
+  void a2a_chunk(const std::vector<LayerView>& views, DType dt,
+                 ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg);
   uint8_t* staging(int64_t bytes, hipStream_t user);     // single-stream path
   uint8_t* slot_bytes(StagingSlot& slot, int64_t bytes); // pipelined path
   void chain(hipStream_t from, hipStream_t to);  // event: `to` waits `from`
   hipEvent_t next_ev();
 
   // Launch one quantize "job list", grouping slices by (bits, bucket%8==0).
-  // fb_base (optional): error-feedback buffer in chunk element space.
+  // fb_base (optional): error-feedback buffer indexed by slice fb_off minus
+  // fb_rebase elements (phase 2 buffers cover only this rank's partition, so
+  // its slices rebase by the partition start — computed host-side per slice,
+  // never by out-of-bounds pointer arithmetic).
   void run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
                     DType dt, hipStream_t stream, bool stochastic,
-                    char* fb_base = nullptr);
-  // per-(chunk, phase) persistent error-feedback buffers
-  at::Tensor& feedback_buf(const void* key_ptr, int64_t numel, int phase,
+                    char* fb_base = nullptr, int64_t fb_rebase = 0);
+  // Persistent error-feedback residuals keyed by (chunk identity, phase).
+  // Bounded: on cap overflow the store resets (residuals are a convergence
+  // aid, losing them once is benign; unbounded growth is not).
+  at::Tensor& feedback_buf(int64_t key, int phase, int64_t numel,
                            at::ScalarType st);
-  std::map<std::tuple<const void*, int64_t, int>, at::Tensor> fb_bufs_;
+  std::map<std::pair<int64_t, int>, at::Tensor> fb_bufs_;
   void run_dequant(const std::vector<Slice>& slices, const uint8_t* in_base,
                    int64_t src_stride, int nsrc, bool add, DType dt,
                    hipStream_t stream);
@@ -241,6 +270,7 @@ class Engine {
   int timer_slot_ = -1;
 
   int rank_, size_;
+  bool is_cross_ = false;
   at::Tensor staging_;
   StagingSlot slots_[2];
   int slot_cur_ = 0;

@@ -387,11 +387,14 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
         }
       }
     } else if (full && ngroups <= MAXG * kWave && fbp) {
-      // error-feedback: stash the T-rounded x+fb and reduce over it
+      // error-feedback: stash the T-rounded x+fb and reduce over it.
+      // fb alignment is independent of the input's (the engine rebases the
+      // phase-2 feedback base by a partition offset), so probe it separately.
+      const bool al16_fb = (reinterpret_cast<uintptr_t>(fbp) & 15) == 0;
       for (int g = lane, gi = 0; g < ngroups; g += kWave, gi++) {
         uint32_t xr[8], fr[8];
         load8<T>(in + g * 8, true, xr);
-        load8<T>(fbp + g * 8, al16, fr);
+        load8<T>(fbp + g * 8, al16_fb, fr);
 #pragma unroll
         for (int j = 0; j < 8; j++) {
           const uint32_t xe = f2raw<T>(raw2f<T>(xr[j]) + raw2f<T>(fr[j]));

@@ -44,10 +44,11 @@ class CGXState:
                 "bits", self.quantization_bits)
             self.quantization_bucket_size = compression_params.get(
                 "bucket_size", self.quantization_bucket_size)
-            if compression_params.get("error_feedback"):
-                # engine config is env-driven (re-read every bucket), so
-                # every rank picks this up consistently
-                os.environ["CGX_ERROR_FEEDBACK"] = "1"
+            # engine config is env-driven (re-read every bucket), so every
+            # rank picks this up consistently; write "0" explicitly when EF
+            # is absent so a later CGXState doesn't inherit a stale "1"
+            os.environ["CGX_ERROR_FEEDBACK"] = (
+                "1" if compression_params.get("error_feedback") else "0")
 
     def should_compress_(self, tensor: torch.Tensor) -> bool:
         if tensor.dim() <= 1 or tensor.numel() < self.layer_min_size:
