@@ -39,6 +39,9 @@ def parse_args():
     p.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
                    help="cpu: harness validation only (gloo delegation)")
     p.add_argument("--channels-last", type=int, default=1)
+    p.add_argument("--show-registry", action="store_true",
+                   help="report the layerwise filter's registry split "
+                        "(compressed vs bits=32 layers) in the JSON config")
     return p.parse_args()
 
 
@@ -63,13 +66,26 @@ def main():
             torch.cuda.synchronize()
 
     bits = 32 if args.no_compress else args.bits
-    distributed = world > 1
-    if distributed:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29500")
-        if args.backend == "cgx":
-            import torch_cgx_amd  # noqa: F401
-        dist.init_process_group(args.backend, rank=rank, world_size=world)
+    # Always initialize the process group and run under DDP + cgx_hook, even
+    # at world_size 1: the backend, bucket traversal, layer registry and hook
+    # are then always in the timed region.  Compression/communication only
+    # occurs at world > 1 — the printed config says so explicitly
+    # ("compression_active") so an N=1 number cannot be read as a
+    # compressed-allreduce measurement.
+    distributed = True
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    if world == 1 and "MASTER_PORT" not in os.environ:
+        # self-rendezvous: bind an ephemeral port so concurrent single-GPU
+        # invocations cannot collide
+        import socket
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            os.environ["MASTER_PORT"] = str(s.getsockname()[1])
+    os.environ.setdefault("MASTER_PORT", "29500")
+    if args.backend == "cgx":
+        import torch_cgx_amd  # noqa: F401
+    dist.init_process_group(args.backend, rank=rank, world_size=world)
+    compression_active = world > 1 and bits <= 8
 
     from torch_cgx_amd.models import resnet50, bert_large
 
@@ -150,6 +166,21 @@ def main():
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = e.item()
 
+    registry_split = None
+    if args.show_registry and args.backend == "cgx":
+        # the hook registered every layer at step 2 (needs >= 3 total steps
+        # incl. warmup): report how the layerwise filter split them
+        # (bias/LayerNorm/small layers -> bits=32, i.e. not compressed)
+        from torch_cgx_amd import _C
+        comp = skip = 0
+        for _idx, _numels, cfgs in _C.registry_snapshot():
+            for b, _bs in cfgs:
+                if b <= 8:
+                    comp += 1
+                else:
+                    skip += 1
+        registry_split = {"compressed_layers": comp, "bits32_layers": skip}
+
     n_gpus = world
     samples_per_sec = batch * n_gpus * args.steps / elapsed
     if rank == 0:
@@ -177,8 +208,14 @@ def main():
                 "backend": args.backend,
                 "bits": bits,
                 "bucket_size": args.bucket_size,
+                # no gradient exchange happens at world_size 1: an N=1 run
+                # is the bf16 compute baseline, not a compressed-allreduce
+                # measurement
+                "compression_active": compression_active,
             },
         }
+        if registry_split is not None:
+            result["config"]["layerwise_filter"] = registry_split
         print(json.dumps(result), flush=True)
 
     if distributed:

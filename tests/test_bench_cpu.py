@@ -33,8 +33,18 @@ def test_bench_resnet_cpu_ws2():
     assert d["value"] > 0
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["bits"] == 4
+    assert d["config"]["compression_active"] is True
     assert d["scaling"] == "weak"
     assert d["data"] == "synthetic"
+
+
+def test_bench_show_registry_ws2():
+    d = _run_bench(["--gpus", "2", "--show-registry"])
+    split = d["config"]["layerwise_filter"]
+    # ResNet-50: conv kernels compressed; BN weights/biases (dim<=1) and
+    # small layers registered at bits=32 by the layerwise filter
+    assert split["compressed_layers"] > 0
+    assert split["bits32_layers"] > 0
 
 
 def test_bench_json_contract_single():
@@ -50,3 +60,6 @@ def test_bench_json_contract_single():
                 "dtype", "data", "config"]:
         assert key in d, key
     assert d["n_gpus"] == 1
+    # an N=1 run must be honestly labeled: the backend+hook run, but no
+    # gradient compression/communication happens
+    assert d["config"]["compression_active"] is False
