@@ -170,6 +170,7 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
     std::memcpy(&id, v.data(), sizeof(id));
   }
   CGX_NCCL_CHECK(ncclCommInitRank(&comm_, size_, id, rank_));
+  tr_ = std::make_unique<RcclTransport>(comm_, rank_, size_);
   stream_ = c10::hip::getStreamFromPoolMasqueradingAsCUDA(
       /*isHighPriority=*/true, device_index_);
   engine_ = std::make_unique<Engine>(rank_, size_);
@@ -199,6 +200,8 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
     }
     CGX_NCCL_CHECK(ncclCommInitRank(&intra_comm_, topo_.local_size, iid,
                                     topo_.local_rank));
+    intra_tr_ = std::make_unique<RcclTransport>(intra_comm_, topo_.local_rank,
+                                                topo_.local_size);
     // cross communicator: peers with the same local_rank on every node
     const std::string ckey =
         "cgx/cross_uid/" + std::to_string(topo_.local_rank);
@@ -214,6 +217,8 @@ void ProcessGroupCGX::lazyInit(at::Device device) {
     }
     CGX_NCCL_CHECK(
         ncclCommInitRank(&cross_comm_, topo_.n_nodes, cid, topo_.node_id));
+    cross_tr_ = std::make_unique<RcclTransport>(cross_comm_, topo_.node_id,
+                                                topo_.n_nodes);
     intra_engine_ = std::make_unique<Engine>(topo_.local_rank,
                                              topo_.local_size);
     cross_engine_ = std::make_unique<Engine>(topo_.node_id, topo_.n_nodes,
@@ -319,13 +324,14 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::allreduce(
         cgx::Registry::BucketInfo info;
         const bool matched =
             cgx::Registry::get().next(t.numel(), t.data_ptr(), &info);
-        fin = intra_engine_->allreduce(t, intra_comm_, stream_->stream(),
-                                       &info, matched);
+        fin = intra_engine_->allreduce(t, intra_tr_.get(),
+                                       stream_->stream(), &info, matched);
         if (topo_.local_rank == 0)
-          fin = cross_engine_->allreduce(t, cross_comm_, fin, &info, matched);
-        fin = intra_engine_->broadcast(t, /*root=*/0, intra_comm_, fin);
+          fin = cross_engine_->allreduce(t, cross_tr_.get(), fin, &info,
+                                         matched);
+        fin = intra_engine_->broadcast(t, /*root=*/0, intra_tr_.get(), fin);
       } else {
-        fin = engine_->allreduce(t, comm_, stream_->stream());
+        fin = engine_->allreduce(t, tr_.get(), stream_->stream());
       }
     }
     auto fin_masq = c10::hip::getStreamFromExternalMasqueradingAsCUDA(

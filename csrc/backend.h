@@ -140,6 +140,7 @@ class ProcessGroupCGX : public c10d::Backend {
   c10::intrusive_ptr<c10d::Store> store_;
   c10::intrusive_ptr<c10d::Backend> cpu_;
   ncclComm_t comm_ = nullptr;
+  std::unique_ptr<RcclTransport> tr_;
   int device_index_ = -1;
   std::unique_ptr<Engine> engine_;
   // hierarchical (multi-node) mode: intra-node SRA + cross-node reduction on
@@ -148,6 +149,8 @@ class ProcessGroupCGX : public c10d::Backend {
   Topology topo_;
   ncclComm_t intra_comm_ = nullptr;
   ncclComm_t cross_comm_ = nullptr;
+  std::unique_ptr<RcclTransport> intra_tr_;
+  std::unique_ptr<RcclTransport> cross_tr_;
   std::unique_ptr<Engine> intra_engine_;
   std::unique_ptr<Engine> cross_engine_;
   bool hierarchical_ = false;

@@ -11,11 +11,92 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/csrc/utils/pybind.h>
 
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+
+#include <thread>
+
 #include "backend.h"
 #include "engine.h"
+#include "transport.h"
 
 namespace cgx {
 namespace {
+
+// Drive the REAL engine code at world size ws on ONE GPU: `buckets[r]` is
+// rank r's flat bucket; ws Engine instances (each with its own streams,
+// staging and events — exactly the production objects) exchange compressed
+// chunks through the loopback transport (device-to-device copies with full
+// stream/event fencing).  This is the hardware test seam the multi-rank
+// SRA/Ring path runs under `pytest -m gpu` with a single MI355X: RCCL
+// itself refuses two ranks on one device.
+void run_loopback(const std::vector<at::Tensor>& tensors,
+                  const std::function<hipStream_t(Engine&, Transport*,
+                                                  hipStream_t, int)>& body) {
+  const int ws = (int)tensors.size();
+  TORCH_CHECK(ws >= 2, "loopback: need >= 2 per-rank tensors");
+  for (const auto& t : tensors) {
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous(),
+                "loopback: CUDA contiguous tensors required");
+    TORCH_CHECK(t.device() == tensors[0].device() &&
+                    t.numel() == tensors[0].numel() &&
+                    t.scalar_type() == tensors[0].scalar_type(),
+                "loopback: tensors must match in device/numel/dtype");
+  }
+  const int dev = tensors[0].device().index();
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(dev);
+  LoopbackHub hub(ws);
+  std::vector<std::unique_ptr<Engine>> engines;
+  std::vector<std::unique_ptr<LoopbackTransport>> trs;
+  std::vector<hipStream_t> qs(ws, nullptr);
+  for (int r = 0; r < ws; r++) {
+    engines.push_back(std::make_unique<Engine>(r, ws));
+    trs.push_back(std::make_unique<LoopbackTransport>(&hub, r));
+    CGX_HIP_CHECK(hipStreamCreateWithFlags(&qs[r], hipStreamNonBlocking));
+  }
+  std::vector<std::exception_ptr> errs(ws);
+  {
+    py::gil_scoped_release nogil;
+    std::vector<std::thread> th;
+    for (int r = 0; r < ws; r++) {
+      th.emplace_back([&, r] {
+        try {
+          c10::hip::HIPGuardMasqueradingAsCUDA g(dev);
+          hipStream_t fin = body(*engines[r], trs[r].get(), qs[r], r);
+          CGX_HIP_CHECK(hipStreamSynchronize(fin));
+          CGX_HIP_CHECK(hipStreamSynchronize(qs[r]));
+          CGX_HIP_CHECK(hipStreamSynchronize(engines[r]->comm_stream()));
+          CGX_HIP_CHECK(hipStreamSynchronize(engines[r]->deq_stream()));
+        } catch (...) {
+          errs[r] = std::current_exception();
+          hub.abort();  // unblock peers so join() cannot hang
+        }
+      });
+    }
+    for (auto& t : th) t.join();
+  }
+  (void)hipDeviceSynchronize();  // loopback events/copies fully retired
+  for (auto s : qs)
+    if (s) (void)hipStreamDestroy(s);
+  for (int r = 0; r < ws; r++)
+    if (errs[r]) std::rethrow_exception(errs[r]);
+}
+
+void py_loopback_allreduce(std::vector<at::Tensor> buckets) {
+  const int64_t numel = buckets.empty() ? 0 : buckets[0].numel();
+  // resolve the registry ONCE and force it into every engine, exactly like
+  // the hierarchical path (one cursor step per bucket across all ranks)
+  Registry::BucketInfo info;
+  const bool matched = Registry::get().next(numel, nullptr, &info);
+  run_loopback(buckets, [&](Engine& e, Transport* tr, hipStream_t qs, int r) {
+    return e.allreduce(buckets[r], tr, qs, &info, matched);
+  });
+}
+
+void py_loopback_broadcast(std::vector<at::Tensor> tensors, int64_t root) {
+  run_loopback(tensors, [&](Engine& e, Transport* tr, hipStream_t qs, int r) {
+    return e.broadcast(tensors[r], (int)root, tr, qs);
+  });
+}
 
 DType dtype_arg(const at::Tensor& t) { return dtype_of(t); }
 
@@ -228,6 +309,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     d["error_feedback"] = c.error_feedback;
     return d;
   });
+  m.def("loopback_allreduce", &cgx::py_loopback_allreduce, py::arg("buckets"),
+        "Run the production Engine::allreduce at world_size=len(buckets) on "
+        "one GPU via the loopback transport; buckets[r] is rank r's flat "
+        "bucket tensor, reduced in place.");
+  m.def("loopback_broadcast", &cgx::py_loopback_broadcast, py::arg("tensors"),
+        py::arg("root") = 0,
+        "Run the production Engine::broadcast at world_size=len(tensors) on "
+        "one GPU via the loopback transport.");
   m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
         py::arg("bucket_size"), py::arg("stochastic") = false,
         py::arg("seed") = 0, py::arg("skip_incomplete") = false,

@@ -539,7 +539,7 @@ void Engine::timer_finish() {
 }
 
 void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
-                       ncclComm_t comm, hipStream_t qs,
+                       Transport* tr, hipStream_t qs,
                        const EngineConfig& cfg, int64_t fb_key) {
   const int ws = size_;
   ChunkPlan pl = plan(views, dt, cfg.skip_incomplete);
@@ -602,17 +602,16 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
 
   // round 1 exchange on the comm stream (grouped p2p drives all xGMI links)
   chain(qs, comm_stream_);
-  CGX_NCCL_CHECK(ncclGroupStart());
-  for (int p = 0; p < ws; p++) {
-    if (p == rank_) continue;
-    if (comp[p] > 0)
-      CGX_NCCL_CHECK(ncclSend(send1 + peer_off[p], comp[p], ncclUint8, p,
-                              comm, comm_stream_));
-    if (mycomp > 0)
-      CGX_NCCL_CHECK(ncclRecv(recv1 + (int64_t)slot(p) * mycomp, mycomp,
-                              ncclUint8, p, comm, comm_stream_));
+  {
+    std::vector<Transport::Op> sends, recvs;
+    for (int p = 0; p < ws; p++) {
+      if (p == rank_) continue;
+      sends.push_back(Transport::Op{send1 + peer_off[p], comp[p], p});
+      recvs.push_back(
+          Transport::Op{recv1 + (int64_t)slot(p) * mycomp, mycomp, p});
+    }
+    tr->exchange(sends, recvs, comm_stream_);
   }
-  CGX_NCCL_CHECK(ncclGroupEnd());
   timer_mark(2, comm_stream_);
 
   // decode-accumulate + self-quantize on the deq stream
@@ -642,17 +641,15 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
 
   // round 2 exchange on the comm stream
   chain(deq_stream_, comm_stream_);
-  CGX_NCCL_CHECK(ncclGroupStart());
-  for (int p = 0; p < ws; p++) {
-    if (p == rank_) continue;
-    if (mycomp > 0)
-      CGX_NCCL_CHECK(
-          ncclSend(send2, mycomp, ncclUint8, p, comm, comm_stream_));
-    if (comp[p] > 0)
-      CGX_NCCL_CHECK(ncclRecv(recv2 + peer_off[p], comp[p], ncclUint8, p,
-                              comm, comm_stream_));
+  {
+    std::vector<Transport::Op> sends, recvs;
+    for (int p = 0; p < ws; p++) {
+      if (p == rank_) continue;
+      sends.push_back(Transport::Op{send2, mycomp, p});
+      recvs.push_back(Transport::Op{recv2 + peer_off[p], comp[p], p});
+    }
+    tr->exchange(sends, recvs, comm_stream_);
   }
-  CGX_NCCL_CHECK(ncclGroupEnd());
   timer_mark(4, comm_stream_);
 
   // final decode: own chunk from send2, peers' chunks from recv2
@@ -678,7 +675,7 @@ void Engine::sra_chunk(const std::vector<LayerView>& views, DType dt,
 }
 
 void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
-                        ncclComm_t comm, hipStream_t stream,
+                        Transport* tr, hipStream_t stream,
                         const EngineConfig& cfg) {
   // Compressed ring allreduce (reference MPI_Allreduce_Ring semantics,
   // ring.cc:139-226): ws-1 reduce-scatter steps with per-hop requantize of
@@ -714,14 +711,8 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
     const int recv_c = (rank_ - s - 1 + ws) % ws;
     if (comp[send_c] > 0)
       run_quantize(rs[send_c], tmp_send, dt, stream, cfg.stochastic);
-    CGX_NCCL_CHECK(ncclGroupStart());
-    if (comp[send_c] > 0)
-      CGX_NCCL_CHECK(
-          ncclSend(tmp_send, comp[send_c], ncclUint8, next, comm, stream));
-    if (comp[recv_c] > 0)
-      CGX_NCCL_CHECK(
-          ncclRecv(tmp_recv, comp[recv_c], ncclUint8, prev, comm, stream));
-    CGX_NCCL_CHECK(ncclGroupEnd());
+    tr->exchange({Transport::Op{tmp_send, comp[send_c], next}},
+                 {Transport::Op{tmp_recv, comp[recv_c], prev}}, stream);
     if (comp[recv_c] > 0)
       run_dequant(rs[recv_c], tmp_recv, 0, 1, /*add=*/true, dt, stream);
   }
@@ -735,14 +726,9 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
   for (int s = 0; s < ws - 1; s++) {
     const int send_c = (own - s + ws) % ws;
     const int recv_c = (own - s - 1 + ws) % ws;
-    CGX_NCCL_CHECK(ncclGroupStart());
-    if (comp[send_c] > 0)
-      CGX_NCCL_CHECK(ncclSend(segs + seg_off[send_c], comp[send_c], ncclUint8,
-                              next, comm, stream));
-    if (comp[recv_c] > 0)
-      CGX_NCCL_CHECK(ncclRecv(segs + seg_off[recv_c], comp[recv_c], ncclUint8,
-                              prev, comm, stream));
-    CGX_NCCL_CHECK(ncclGroupEnd());
+    tr->exchange({Transport::Op{segs + seg_off[send_c], comp[send_c], next}},
+                 {Transport::Op{segs + seg_off[recv_c], comp[recv_c], prev}},
+                 stream);
   }
 
   // final decode of every segment (own included, for bit-identical results)
@@ -758,7 +744,7 @@ void Engine::ring_chunk(const std::vector<LayerView>& views, DType dt,
 }
 
 void Engine::a2a_chunk(const std::vector<LayerView>& views, DType dt,
-                       ncclComm_t comm, hipStream_t qs,
+                       Transport* tr, hipStream_t qs,
                        const EngineConfig& cfg) {
   // Debug brute-force reduction (reference AllReduceAlltoAllCompressed,
   // scatter_reduce_allgather.cc:269-306): every rank quantizes its ENTIRE
@@ -787,15 +773,16 @@ void Engine::a2a_chunk(const std::vector<LayerView>& views, DType dt,
   uint8_t* recv = base + C;      // ws-1 peer streams, stride C
   run_quantize(sl, send, dt, qs, cfg.stochastic);
   chain(qs, comm_stream_);
-  CGX_NCCL_CHECK(ncclGroupStart());
-  for (int p = 0; p < ws; p++) {
-    if (p == rank_) continue;
-    const int s = p < rank_ ? p : p - 1;
-    CGX_NCCL_CHECK(ncclSend(send, C, ncclUint8, p, comm, comm_stream_));
-    CGX_NCCL_CHECK(ncclRecv(recv + (int64_t)s * C, C, ncclUint8, p, comm,
-                            comm_stream_));
+  {
+    std::vector<Transport::Op> sends, recvs;
+    for (int p = 0; p < ws; p++) {
+      if (p == rank_) continue;
+      const int s = p < rank_ ? p : p - 1;
+      sends.push_back(Transport::Op{send, C, p});
+      recvs.push_back(Transport::Op{recv + (int64_t)s * C, C, p});
+    }
+    tr->exchange(sends, recvs, comm_stream_);
   }
-  CGX_NCCL_CHECK(ncclGroupEnd());
   chain(comm_stream_, deq_stream_);
   run_dequant(sl, send, 0, 1, /*add=*/false, dt, deq_stream_);
   if (ws > 1)
@@ -805,7 +792,7 @@ void Engine::a2a_chunk(const std::vector<LayerView>& views, DType dt,
   chain(deq_stream_, qs);
 }
 
-hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
+hipStream_t Engine::broadcast(at::Tensor t, int root, Transport* tr,
                               hipStream_t qs) {
   const EngineConfig cfg = EngineConfig::from_env();
   const int64_t n = t.numel();
@@ -816,8 +803,8 @@ hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
                          t.scalar_type() == at::kBFloat16);
   if (!compress || size_ <= 1) {
     if (size_ > 1) {
-      CGX_NCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), n,
-                                   nccl_dtype(t), root, comm, qs));
+      tr->broadcast(t.data_ptr(), n, (int)t.element_size(), nccl_dtype(t),
+                    root, qs);
     }
     return qs;
   }
@@ -832,13 +819,13 @@ hipStream_t Engine::broadcast(at::Tensor t, int root, ncclComm_t comm,
   if (rank_ == root) {
     run_quantize(sl, buf, dt, qs, cfg.stochastic);
   }
-  CGX_NCCL_CHECK(ncclBroadcast(buf, buf, bytes, ncclUint8, root, comm, qs));
+  tr->broadcast(buf, bytes, 1, ncclUint8, root, qs);
   // every rank (root included) decodes the same bytes -> bit-identical
   run_dequant(sl, buf, 0, 1, /*add=*/false, dt, qs);
   return qs;
 }
 
-hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
+hipStream_t Engine::allreduce(at::Tensor bucket, Transport* tr,
                               hipStream_t qs,
                               const Registry::BucketInfo* forced,
                               bool forced_match) {
@@ -892,12 +879,9 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
   if (!uncomp.empty()) {
     const ncclDataType_t ndt = nccl_dtype(bucket);
     chain(qs, comm_stream_);
-    CGX_NCCL_CHECK(ncclGroupStart());
-    for (auto& [ptr, cnt] : uncomp) {
-      CGX_NCCL_CHECK(
-          ncclAllReduce(ptr, ptr, cnt, ndt, ncclSum, comm, comm_stream_));
-    }
-    CGX_NCCL_CHECK(ncclGroupEnd());
+    std::vector<std::pair<void*, int64_t>> bufs;
+    for (auto& [ptr, cnt] : uncomp) bufs.emplace_back(ptr, cnt);
+    tr->allreduce_sum_group(bufs, ndt, comm_stream_);
   }
 
   // tensor-fusion chunking over the compressible layers
@@ -937,17 +921,17 @@ hipStream_t Engine::allreduce(at::Tensor bucket, ncclComm_t comm,
     }
     const std::vector<LayerView>& vs = *vsp;
     if (cfg.debug_a2a) {
-      a2a_chunk(vs, dt, comm, qs, cfg);
+      a2a_chunk(vs, dt, tr, qs, cfg);
     } else if (cfg.ring && size_ > 2) {
       // ring is hop-serial and runs on one stream (qs).  Its NCCL ops must
       // be ordered behind any uncompressed ncclAllReduce group queued on
       // comm_stream_ above: unordered concurrent ops on one communicator
       // are UB in NCCL/RCCL.
       chain(comm_stream_, qs);
-      ring_chunk(vs, dt, comm, qs, cfg);
+      ring_chunk(vs, dt, tr, qs, cfg);
       chain(qs, deq_stream_);             // keep completion on deq stream
     } else {
-      sra_chunk(vs, dt, comm, qs, cfg, fb_key);
+      sra_chunk(vs, dt, tr, qs, cfg, fb_key);
     }
   };
   auto flush = [&]() {

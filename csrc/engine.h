@@ -14,6 +14,8 @@
 #include <ATen/ATen.h>
 #include <rccl/rccl.h>
 
+#include "transport.h"
+
 #include <cstdint>
 #include <map>
 #include <mutex>
@@ -158,7 +160,7 @@ class Engine {
   // registry ONCE per bucket and passes it to both the intra and cross
   // engines -- each engine consuming a registry cursor step would
   // desynchronize leader vs non-leader ranks when bucket totals repeat).
-  hipStream_t allreduce(at::Tensor bucket, ncclComm_t comm, hipStream_t qs,
+  hipStream_t allreduce(at::Tensor bucket, Transport* tr, hipStream_t qs,
                         const Registry::BucketInfo* forced = nullptr,
                         bool forced_match = false);
 
@@ -171,7 +173,7 @@ class Engine {
   // back to plain ncclBroadcast when compression is off.  Used by the
   // hierarchical intra-node broadcast when CGX_INTRA_COMPRESS=1 (default,
   // matching the reference, mpi_allreduce_operations.cc:134).
-  hipStream_t broadcast(at::Tensor t, int root, ncclComm_t comm,
+  hipStream_t broadcast(at::Tensor t, int root, Transport* tr,
                         hipStream_t qs);
 
   // Mirror of the partition walk (reference Quantizer::GetSizesAndOffsets,
@@ -213,10 +215,10 @@ class Engine {
   // residual store — (bucket_idx << 20 | chunk ordinal) when the registry
   // matched, else a pointer-derived fallback key.
   void sra_chunk(const std::vector<LayerView>& views, DType dt,
-                 ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg,
+                 Transport* tr, hipStream_t qs, const EngineConfig& cfg,
                  int64_t fb_key);
   void ring_chunk(const std::vector<LayerView>& views, DType dt,
-                  ncclComm_t comm, hipStream_t stream,
+                  Transport* tr, hipStream_t stream,
                   const EngineConfig& cfg);
   // Brute-force debug reduction (CGX_DEBUG_ALL_TO_ALL_REDUCTION parity,
   // reference scatter_reduce_allgather.cc:269-306): every rank quantizes its
@@ -224,7 +226,7 @@ class Engine {
   // same ws compressed streams — no partitioning, ws× the wire traffic, but
   // removes the partition/offset machinery from the fault surface.
   void a2a_chunk(const std::vector<LayerView>& views, DType dt,
-                 ncclComm_t comm, hipStream_t qs, const EngineConfig& cfg);
+                 Transport* tr, hipStream_t qs, const EngineConfig& cfg);
   uint8_t* staging(int64_t bytes, hipStream_t user);     // single-stream path
   uint8_t* slot_bytes(StagingSlot& slot, int64_t bytes); // pipelined path
   void chain(hipStream_t from, hipStream_t to);  // event: `to` waits `from`

@@ -17,10 +17,12 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SOURCES = [
     "csrc/quant_kernels.hip",
     "csrc/engine.cc",
+    "csrc/transport.cc",
     "csrc/backend.cc",
     "csrc/bindings.cc",
 ]
-HEADERS = ["csrc/compress.h", "csrc/engine.h", "csrc/backend.h"]
+HEADERS = ["csrc/compress.h", "csrc/engine.h", "csrc/backend.h",
+           "csrc/transport.h"]
 EXT_OUT = os.path.join(
     "torch_cgx_amd", "_C" + sysconfig.get_config_var("EXT_SUFFIX"))
 
