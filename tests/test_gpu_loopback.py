@@ -1,0 +1,292 @@
+"""Multi-rank compressed allreduce on ONE MI355X: the production
+Engine::sra_chunk / ring_chunk / a2a_chunk code (C++, real kernels, real
+streams/events/staging) executes at world_size 2..8 through the loopback
+transport and is compared bitwise against the CPU golden simulation.
+
+This is the hardware test seam for the multi-rank path (VERDICT round 1,
+item 1): RCCL refuses two ranks on one device, so `_C.loopback_allreduce`
+spawns N engine instances inside one process that exchange compressed
+chunks via device-to-device copies with full stream/event fencing — the
+exact orchestration code that runs over RCCL in production
+(csrc/engine.cc, csrc/transport.cc).
+"""
+
+import os
+
+import pytest
+import torch
+
+from torch_cgx_amd.ops import golden
+from torch_cgx_amd.parallel import sra_sim
+
+pytestmark = pytest.mark.gpu
+
+
+def _dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda:0")
+
+
+ENV_KEYS = (
+    "CGX_COMPRESSION_QUANTIZATION_BITS", "CGX_COMPRESSION_BUCKET_SIZE",
+    "CGX_STOCHASTIC_ROUNDING", "CGX_INNER_REDUCTION_TYPE",
+    "CGX_CROSS_REDUCTION_TYPE", "CGX_REDUCTION_TYPE",
+    "CGX_FUSION_BUFFER_SIZE_MB", "CGX_DEBUG_ALL_TO_ALL_REDUCTION",
+    "CGX_ERROR_FEEDBACK", "CGX_COMPRESSION_MINIMAL_SIZE",
+)
+
+
+@pytest.fixture(autouse=True)
+def _env():
+    saved = {k: os.environ.get(k) for k in ENV_KEYS}
+    from torch_cgx_amd import _C
+    _C.clear_registry()
+    yield os.environ
+    for k, v in saved.items():
+        if v is None:
+            os.environ.pop(k, None)
+        else:
+            os.environ[k] = v
+    _C.clear_registry()
+
+
+def _cfg(env, bits, bucket, stochastic=False):
+    env["CGX_COMPRESSION_QUANTIZATION_BITS"] = str(bits)
+    env["CGX_COMPRESSION_BUCKET_SIZE"] = str(bucket)
+    env["CGX_STOCHASTIC_ROUNDING"] = "1" if stochastic else "0"
+
+
+@pytest.mark.parametrize("ws", [2, 4, 8])
+@pytest.mark.parametrize("bits", [2, 4, 8])
+@pytest.mark.parametrize("dtype",
+                         [torch.float32, torch.float16, torch.bfloat16])
+def test_sra_equal_inputs_exact(_env, ws, bits, dtype):
+    """All ranks hold identical data -> max-min quantization is exact and
+    the reduced result must equal ws * x exactly (reference
+    test_compressed_exact parity at real multi-rank)."""
+    from torch_cgx_amd import _C
+    _cfg(_env, bits, 512)
+    for n in [128, 1024, 100_000]:
+        buckets = [torch.full((n,), 3.0, dtype=dtype, device=_dev())
+                   for _ in range(ws)]
+        _C.loopback_allreduce(buckets)
+        expected = torch.full((n,), 3.0 * ws, dtype=dtype, device=_dev())
+        for r, b in enumerate(buckets):
+            assert torch.equal(b, expected), (ws, bits, dtype, n, r)
+
+
+@pytest.mark.parametrize("ws", [2, 3, 4, 8])
+@pytest.mark.parametrize("bits,bucket", [(4, 512), (8, 64), (2, 1024)])
+def test_sra_matches_cpu_sim_bitwise(_env, ws, bits, bucket):
+    """Random inputs, deterministic rounding: the hardware multi-rank SRA
+    must produce byte-for-byte the result of the CPU golden simulation."""
+    from torch_cgx_amd import _C
+    _cfg(_env, bits, bucket)
+    torch.manual_seed(100 + ws)
+    n = 40_000
+    cpu = [torch.randn(n) for _ in range(ws)]
+    sim = sra_sim.sra_allreduce([t.clone() for t in cpu], [n],
+                                [(bits, bucket)])
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    for r in range(ws):
+        got = buckets[r].cpu()
+        assert torch.equal(got, sim[r]), (
+            ws, bits, bucket, r,
+            (got - sim[r]).abs().max().item())
+    # and all ranks bit-identical, the SRA invariant
+    for r in range(1, ws):
+        assert torch.equal(buckets[r], buckets[0])
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+@pytest.mark.parametrize("bits", [2, 4, 8])
+def test_sra_error_bound_hw(_env, ws, bits):
+    """The reference acceptance suite's analytic bound (test_cgx.py:92), on
+    hardware at real world sizes, with stochastic rounding ON."""
+    from torch_cgx_amd import _C
+    bucket = 512
+    _cfg(_env, bits, bucket, stochastic=True)
+    n = 100_000
+    arange = torch.arange(-n / 2, n / 2, 1.0, dtype=torch.float32)
+    buckets = [((r + 1) * arange).to(_dev()) for r in range(ws)]
+    expected = (ws * (ws + 1) / 2) * arange
+    _C.loopback_allreduce(buckets)
+    bound = 2 * min(bucket, n) / ((1 << bits) - 1) * ws * (ws + 1)
+    for b in buckets:
+        err = (b.cpu() - expected).abs().max().item()
+        assert err < bound, (ws, bits, err, bound)
+
+
+@pytest.mark.parametrize("ws", [3, 4, 8])
+def test_ring_matches_cpu_sim_bitwise(_env, ws):
+    """CGX_INNER_REDUCTION_TYPE=Ring at ws>2: the hardware ring (per-hop
+    requantize of running partial sums + allgather forwarding) must match
+    the CPU ring simulation byte for byte."""
+    from torch_cgx_amd import _C
+    bits, bucket = 4, 512
+    _cfg(_env, bits, bucket)
+    _env["CGX_INNER_REDUCTION_TYPE"] = "Ring"
+    torch.manual_seed(7 + ws)
+    n = 30_000
+    cpu = [torch.randn(n) for _ in range(ws)]
+    sim = sra_sim.ring_allreduce([t.clone() for t in cpu], [n],
+                                 [(bits, bucket)])
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    for r in range(ws):
+        assert torch.equal(buckets[r].cpu(), sim[r]), (ws, r)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_sra_multichunk_fusion(_env, ws):
+    """Tiny fusion threshold -> many chunks through the double-buffered
+    staging slots and the descriptor ring; per-chunk partitioning must match
+    a per-chunk CPU simulation."""
+    from torch_cgx_amd import _C
+    bits, bucket = 4, 512
+    _cfg(_env, bits, bucket)
+    _env["CGX_FUSION_BUFFER_SIZE_MB"] = "0"  # clamps to 2048 B = 512 fp32
+    fusion_elems = 512
+    torch.manual_seed(21)
+    n = 10_000  # 20 chunks, incl. a ragged tail
+    cpu = [torch.randn(n) for _ in range(ws)]
+    expected = [torch.empty(n) for _ in range(ws)]
+    for o in range(0, n, fusion_elems):
+        m = min(fusion_elems, n - o)
+        piece = sra_sim.sra_allreduce([t[o:o + m].clone() for t in cpu],
+                                      [m], [(bits, bucket)])
+        for r in range(ws):
+            expected[r][o:o + m] = piece[r]
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    for r in range(ws):
+        assert torch.equal(buckets[r].cpu(), expected[r]), (ws, r)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_registered_mixed_layers(_env, ws):
+    """A registered bucket with mixed per-layer configs, including a bits=32
+    layer that must travel uncompressed (loopback allreduce_sum) while its
+    compressed neighbors go through SRA."""
+    from torch_cgx_amd import _C
+    _cfg(_env, 32, 512)  # defaults say no compression; registry overrides
+    layers = [1000, 513, 2048]
+    cfgs = [(4, 512), (32, 512), (8, 64)]
+    for li, (numel, (b, bs)) in enumerate(zip(layers, cfgs)):
+        _C.register_layer(0, li, numel, b, bs)
+    torch.manual_seed(3)
+    n = sum(layers)
+    cpu = [torch.randn(n) for _ in range(ws)]
+
+    # expected: compressed layers [0] and [2] form the SRA view list;
+    # layer 1 is an exact float sum
+    comp_cat = [torch.cat([t[:1000], t[1513:]]) for t in cpu]
+    sim = sra_sim.sra_allreduce([t.clone() for t in comp_cat],
+                                [1000, 2048], [(4, 512), (8, 64)])
+    exact_mid = sum(t[1000:1513] for t in cpu)
+
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    for r in range(ws):
+        got = buckets[r].cpu()
+        assert torch.equal(got[:1000], sim[r][:1000]), ("layer0", r)
+        assert torch.equal(got[1513:], sim[r][1000:]), ("layer2", r)
+        assert torch.allclose(got[1000:1513], exact_mid, atol=1e-4), (
+            "uncompressed layer", r)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_debug_a2a_reduction(_env, ws):
+    """CGX_DEBUG_ALL_TO_ALL_REDUCTION: equal inputs reduce exactly; random
+    inputs land within the single-quantization bound (each rank's data is
+    quantized exactly once)."""
+    from torch_cgx_amd import _C
+    bits, bucket = 4, 512
+    _cfg(_env, bits, bucket)
+    _env["CGX_DEBUG_ALL_TO_ALL_REDUCTION"] = "1"
+    n = 50_000
+    buckets = [torch.full((n,), 2.0, device=_dev()) for _ in range(ws)]
+    _C.loopback_allreduce(buckets)
+    for b in buckets:
+        assert torch.equal(b, torch.full((n,), 2.0 * ws, device=_dev()))
+
+    torch.manual_seed(11)
+    cpu = [torch.randn(n) for _ in range(ws)]
+    expected = sum(cpu)
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    bound = 2 * bucket / ((1 << bits) - 1) * ws  # ws once-quantized streams
+    for b in buckets:
+        err = (b.cpu() - expected).abs().max().item()
+        assert err < bound, (err, bound)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_error_feedback_multirank(_env, ws):
+    """CGX_ERROR_FEEDBACK at real multi-rank: executes the phase-1 and
+    (rebased) phase-2 feedback paths; the result stays within the analytic
+    bound and all ranks agree bitwise."""
+    from torch_cgx_amd import _C
+    bits, bucket = 4, 512
+    _cfg(_env, bits, bucket)
+    _env["CGX_ERROR_FEEDBACK"] = "1"
+    torch.manual_seed(5)
+    n = 20_000
+    cpu = [torch.randn(n) for _ in range(ws)]
+    expected = sum(cpu)
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    bound = 2 * bucket / ((1 << bits) - 1) * ws * (ws + 1)
+    for r in range(ws):
+        err = (buckets[r].cpu() - expected).abs().max().item()
+        assert err < bound, (r, err, bound)
+    for r in range(1, ws):
+        assert torch.equal(buckets[r], buckets[0])
+
+
+def test_ring_error_feedback_raises(_env):
+    from torch_cgx_amd import _C
+    _cfg(_env, 4, 512)
+    _env["CGX_INNER_REDUCTION_TYPE"] = "Ring"
+    _env["CGX_ERROR_FEEDBACK"] = "1"
+    buckets = [torch.randn(4096, device=_dev()) for _ in range(3)]
+    with pytest.raises(RuntimeError, match="ERROR_FEEDBACK"):
+        _C.loopback_allreduce(buckets)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_compressed_broadcast(_env, ws):
+    """Engine::broadcast at multi-rank: every rank must end with the
+    root's quantize->dequantize round trip, bit-identical (root included)."""
+    from torch_cgx_amd import _C
+    bits, bucket = 8, 512
+    _cfg(_env, bits, bucket)
+    torch.manual_seed(9)
+    n = 10_000
+    root_cpu = torch.randn(n)
+    expected = golden.dequantize(
+        golden.quantize(root_cpu, bits, bucket, rand=0.5), n,
+        torch.float32, bits, bucket)
+    tensors = [root_cpu.to(_dev()) if r == 0
+               else torch.zeros(n, device=_dev()) for r in range(ws)]
+    _C.loopback_broadcast(tensors, 0)
+    for r in range(ws):
+        assert torch.equal(tensors[r].cpu(), expected), r
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_sra_low_precision_bitwise(_env, dtype):
+    """fp16/bf16 buckets through the real multi-rank path vs the CPU sim."""
+    from torch_cgx_amd import _C
+    ws, bits, bucket = 4, 4, 512
+    _cfg(_env, bits, bucket)
+    torch.manual_seed(13)
+    n = 16_384
+    cpu = [torch.randn(n).to(dtype) for _ in range(ws)]
+    sim = sra_sim.sra_allreduce([t.clone() for t in cpu], [n],
+                                [(bits, bucket)])
+    buckets = [t.to(_dev()) for t in cpu]
+    _C.loopback_allreduce(buckets)
+    for r in range(ws):
+        assert torch.equal(buckets[r].cpu(), sim[r]), (dtype, r)
