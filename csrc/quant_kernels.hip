@@ -641,17 +641,21 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       // 3x slower per element than the fp16 path's single int4)
       const int64_t full_subs = nq >> 2;
       const uint32_t B4 = oneb ? static_cast<uint32_t>(d.bucket >> 2) : 1u;
-      // two grid-stride iterations in flight: independent decode chains for
-      // latency hiding while every store stays a coalesced float4
+      // four grid-stride iterations in flight: the decode chain is
+      // load(packed)+load(meta)->decode->store, so per-lane outstanding
+      // loads scale with the unroll depth (round-1 profile showed the
+      // 2-deep variant latency-bound at ~2.1 TB/s)
+      constexpr int U = 4;
       int64_t w = t0;
-      for (; w + stride < full_subs; w += 2 * stride) {
-        const int64_t ws2[2] = {w, w + stride};
-        uint32_t v[2][4];
+      for (; w + (U - 1) * stride < full_subs; w += U * stride) {
+        int64_t ws2[U];
+        uint32_t v[U][4];
         const bool have = d.add != 0;
-        float* outp[2];
-        uint32_t bk0[2];
+        float* outp[U];
+        uint32_t bk0[U];
 #pragma unroll
-        for (int u = 0; u < 2; u++) {
+        for (int u = 0; u < U; u++) {
+          ws2[u] = w + u * stride;
           outp[u] = reinterpret_cast<float*>(d.out) + ws2[u] * 4;
           if (have) load4f(outp[u], al16, v[u]);
           bk0[u] = oneb
@@ -663,21 +667,24 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
         for (int sidx = 0; sidx < d.nsrc; sidx++) {
           const uint8_t* src = in0 + sidx * d.src_stride;
           const R* meta = reinterpret_cast<const R*>(src - meta_bytes);
+          uint64_t value[U];
 #pragma unroll
-          for (int u = 0; u < 2; u++) {
+          for (int u = 0; u < U; u++) {
             const int64_t g = ws2[u] >> 1;
             const int h = static_cast<int>(ws2[u] & 1);
-            uint64_t value;
             if constexpr ((BITS & 1) == 0) {
-              value = load_bytes(src + g * BITS + h * (BITS / 2), BITS / 2);
+              value[u] = load_bytes(src + g * BITS + h * (BITS / 2), BITS / 2);
             } else {
-              value = load_bytes(src + g * BITS, BITS) >> (h * 4 * BITS);
+              value[u] = load_bytes(src + g * BITS, BITS) >> (h * 4 * BITS);
             }
+          }
+#pragma unroll
+          for (int u = 0; u < U; u++) {
 #pragma unroll
             for (int j = 0; j < 4; j++) {
               const int64_t bk = oneb ? bk0[u] : (ws2[u] * 4 + j) / d.bucket;
               const uint32_t lvl = static_cast<uint32_t>(
-                  (value >> (j * BITS)) & ((1u << BITS) - 1));
+                  (value[u] >> (j * BITS)) & ((1u << BITS) - 1));
               const uint32_t prod =
                   f2raw<T>(raw2f<T>(meta[2 * bk]) * static_cast<float>(lvl));
               const uint32_t dec =
@@ -690,8 +697,8 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
             }
           }
         }
-        store4f(outp[0], al16, v[0]);
-        store4f(outp[1], al16, v[1]);
+#pragma unroll
+        for (int u = 0; u < U; u++) store4f(outp[u], al16, v[u]);
       }
       for (; w < full_subs; w += stride) {
         const int64_t g = w >> 1;
