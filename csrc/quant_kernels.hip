@@ -22,6 +22,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <cstdlib>
 #include <type_traits>
 
 #include "compress.h"
@@ -599,6 +600,251 @@ __device__ __forceinline__ void load_meta_pair(const R* meta, int64_t bk,
   }
 }
 
+// Branch-free packed loads for the hot BITS values.  The compressed stream
+// base is always 8-byte aligned (align8 slice sizes + 2*sizeof(R)*nb meta),
+// so the typed loads below are aligned by construction.
+// half: 4 elements (fp32 path; h selects the half-pack).
+template <int BITS>
+__device__ __forceinline__ uint64_t load_pack_half(
+    const uint8_t* __restrict__ src, int64_t g, int h) {
+  if constexpr (BITS == 1) {
+    return static_cast<uint64_t>(src[g]) >> (h * 4);
+  } else if constexpr (BITS == 2) {
+    return src[g * 2 + h];
+  } else if constexpr (BITS == 4) {
+    return *reinterpret_cast<const uint16_t*>(src + g * 4 + h * 2);
+  } else if constexpr (BITS == 8) {
+    return *reinterpret_cast<const uint32_t*>(src + g * 8 + h * 4);
+  } else if constexpr ((BITS & 1) == 0) {
+    return load_bytes(src + g * BITS + h * (BITS / 2), BITS / 2);
+  } else {
+    return load_bytes(src + g * BITS, BITS) >> (h * 4 * BITS);
+  }
+}
+
+// full: 8 elements (16-bit dtype path).
+template <int BITS>
+__device__ __forceinline__ uint64_t load_pack_full(
+    const uint8_t* __restrict__ src, int64_t g) {
+  if constexpr (BITS == 1) {
+    return src[g];
+  } else if constexpr (BITS == 2) {
+    return *reinterpret_cast<const uint16_t*>(src + g * 2);
+  } else if constexpr (BITS == 4) {
+    return *reinterpret_cast<const uint32_t*>(src + g * 4);
+  } else if constexpr (BITS == 8) {
+    return *reinterpret_cast<const uint64_t*>(src + g * 8);
+  } else {
+    return load_bytes(src + g * BITS, BITS);
+  }
+}
+
+using v4u = uint32_t __attribute__((ext_vector_type(4)));
+
+// Compile-time-aligned 16-byte load/store (the runtime-bool forms compile
+// into per-dword flat accesses inside the hot loop).
+template <bool AL16>
+__device__ __forceinline__ void load16B(const void* p, uint32_t (&r)[4]) {
+  if constexpr (AL16) {
+    const v4u a = *reinterpret_cast<const v4u*>(
+        __builtin_assume_aligned(p, 16));
+    r[0] = a.x; r[1] = a.y; r[2] = a.z; r[3] = a.w;
+  } else {
+    const uint32_t* q = reinterpret_cast<const uint32_t*>(p);
+#pragma unroll
+    for (int j = 0; j < 4; j++) r[j] = q[j];
+  }
+}
+
+template <bool AL16>
+__device__ __forceinline__ void store16B(void* p, const uint32_t (&r)[4]) {
+  if constexpr (AL16) {
+    v4u a;
+    a.x = r[0]; a.y = r[1]; a.z = r[2]; a.w = r[3];
+    *reinterpret_cast<v4u*>(__builtin_assume_aligned(p, 16)) = a;
+  } else {
+    uint32_t* q = reinterpret_cast<uint32_t*>(p);
+#pragma unroll
+    for (int j = 0; j < 4; j++) q[j] = r[j];
+  }
+}
+
+// Incremental bucket-index tracker: replaces the per-iteration integer
+// division (a ~40-instruction software sequence that dominated the round-1
+// hot loop) with one uniform quotient add + carry per iteration.  Valid
+// while indexes fit in u32 (`small` slices) and the per-iteration step is
+// uniform.
+struct BkTrack {
+  uint32_t bk, rem, q, r, div;
+  __device__ __forceinline__ void init(uint32_t x, uint32_t step,
+                                       uint32_t d) {
+    div = d;
+    bk = x / d;
+    rem = x % d;
+    q = step / d;
+    r = step % d;
+  }
+  __device__ __forceinline__ void advance() {
+    bk += q;
+    rem += r;
+    if (rem >= div) {
+      rem -= div;
+      bk += 1;
+    }
+  }
+};
+
+// fp32 fast path: bucket % 8 == 0, u32-indexable slice.  4 elems/lane, two
+// independent chains, no division, no per-access branches.
+template <int BITS, bool AL16, bool HAVE>
+__device__ void deq_f32_fast(const DequantDesc& d,
+                             const uint8_t* __restrict__ in0,
+                             int64_t meta_bytes, int64_t full_subs,
+                             uint32_t B4, int64_t t0, int64_t stride) {
+  constexpr int U = 2;
+  BkTrack bk[U];
+#pragma unroll
+  for (int u = 0; u < U; u++)
+    bk[u].init(static_cast<uint32_t>(t0 + u * stride),
+               static_cast<uint32_t>(U * stride), B4);
+  int64_t w = t0;
+  for (; w + stride < full_subs; w += U * stride) {
+    uint32_t v[U][4];
+    float* outp[U];
+#pragma unroll
+    for (int u = 0; u < U; u++) {
+      outp[u] = reinterpret_cast<float*>(d.out) + (w + u * stride) * 4;
+      if constexpr (HAVE) load16B<AL16>(outp[u], v[u]);
+    }
+    for (int sidx = 0; sidx < d.nsrc; sidx++) {
+      const uint8_t* __restrict__ src = in0 + sidx * d.src_stride;
+      const float* __restrict__ meta =
+          reinterpret_cast<const float*>(src - meta_bytes);
+      uint64_t value[U];
+      uint32_t ur[U], mr[U];
+#pragma unroll
+      for (int u = 0; u < U; u++) {
+        const int64_t ws = w + u * stride;
+        value[u] = load_pack_half<BITS>(src, ws >> 1,
+                                        static_cast<int>(ws & 1));
+        load_meta_pair<uint32_t>(reinterpret_cast<const uint32_t*>(meta),
+                                 bk[u].bk, ur[u], mr[u]);
+      }
+#pragma unroll
+      for (int u = 0; u < U; u++) {
+        const float unitf = bitcast<float>(ur[u]);
+        const float minf = bitcast<float>(mr[u]);
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          const uint32_t lvl = static_cast<uint32_t>(
+              (value[u] >> (j * BITS)) & ((1u << BITS) - 1));
+          const float dec = minf + unitf * static_cast<float>(lvl);
+          if (!HAVE && sidx == 0) {
+            v[u][j] = bitcast<uint32_t>(dec);
+          } else {
+            v[u][j] = bitcast<uint32_t>(bitcast<float>(v[u][j]) + dec);
+          }
+        }
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < U; u++) {
+      store16B<AL16>(outp[u], v[u]);
+      bk[u].advance();
+    }
+  }
+  // at most one trailing sub per thread
+  for (; w < full_subs; w += stride) {
+    uint32_t v[4];
+    float* outp = reinterpret_cast<float*>(d.out) + w * 4;
+    if constexpr (HAVE) load16B<AL16>(outp, v);
+    const uint32_t bkw = static_cast<uint32_t>(w) / B4;
+    for (int sidx = 0; sidx < d.nsrc; sidx++) {
+      const uint8_t* __restrict__ src = in0 + sidx * d.src_stride;
+      const uint32_t* __restrict__ meta =
+          reinterpret_cast<const uint32_t*>(src - meta_bytes);
+      const uint64_t value =
+          load_pack_half<BITS>(src, w >> 1, static_cast<int>(w & 1));
+      uint32_t ur, mr;
+      load_meta_pair<uint32_t>(meta, bkw, ur, mr);
+      const float unitf = bitcast<float>(ur);
+      const float minf = bitcast<float>(mr);
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        const uint32_t lvl = static_cast<uint32_t>((value >> (j * BITS)) &
+                                                   ((1u << BITS) - 1));
+        const float dec = minf + unitf * static_cast<float>(lvl);
+        if (!HAVE && sidx == 0) {
+          v[j] = bitcast<uint32_t>(dec);
+        } else {
+          v[j] = bitcast<uint32_t>(bitcast<float>(v[j]) + dec);
+        }
+      }
+    }
+    store16B<AL16>(outp, v);
+  }
+}
+
+// 16-bit fast path: 8 elems/lane (one int4 store), incremental bucket index.
+template <typename T, int BITS, bool AL16, bool HAVE>
+__device__ void deq_16_fast(const DequantDesc& d,
+                            const uint8_t* __restrict__ in0,
+                            int64_t meta_bytes, int64_t full_groups,
+                            uint32_t B8, int64_t t0, int64_t stride) {
+  using R = typename RawOf<T>::type;
+  BkTrack bk;
+  bk.init(static_cast<uint32_t>(t0), static_cast<uint32_t>(stride), B8);
+  for (int64_t g = t0; g < full_groups; g += stride) {
+    uint32_t w4[4];
+    T* outp = reinterpret_cast<T*>(d.out) + g * 8;
+    uint32_t v[8];
+    if constexpr (HAVE) {
+      if constexpr (AL16) {
+        load16B<true>(outp, w4);
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          v[j] = (w4[j >> 1] >> ((j & 1) * 16)) & 0xFFFF;
+      } else {
+        // 2-byte-aligned output: element loads (u32 would be misaligned)
+        const R* q = reinterpret_cast<const R*>(outp);
+#pragma unroll
+        for (int j = 0; j < 8; j++) v[j] = q[j];
+      }
+    }
+    for (int sidx = 0; sidx < d.nsrc; sidx++) {
+      const uint8_t* __restrict__ src = in0 + sidx * d.src_stride;
+      const R* __restrict__ meta = reinterpret_cast<const R*>(src - meta_bytes);
+      const uint64_t value = load_pack_full<BITS>(src, g);
+      uint32_t ur, mr;
+      load_meta_pair<R>(meta, bk.bk, ur, mr);
+      const float unitf = raw2f<T>(ur);
+      const float minf = raw2f<T>(mr);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const uint32_t lvl = static_cast<uint32_t>((value >> (j * BITS)) &
+                                                   ((1u << BITS) - 1));
+        const uint32_t prod = f2raw<T>(unitf * static_cast<float>(lvl));
+        const uint32_t dec = f2raw<T>(minf + raw2f<T>(prod));
+        if (!HAVE && sidx == 0) {
+          v[j] = dec;
+        } else {
+          v[j] = f2raw<T>(raw2f<T>(v[j]) + raw2f<T>(dec));
+        }
+      }
+    }
+    if constexpr (AL16) {
+#pragma unroll
+      for (int j = 0; j < 4; j++) w4[j] = v[2 * j] | (v[2 * j + 1] << 16);
+      store16B<true>(outp, w4);
+    } else {
+      R* q = reinterpret_cast<R*>(outp);
+#pragma unroll
+      for (int j = 0; j < 8; j++) q[j] = static_cast<R>(v[j]);
+    }
+    bk.advance();
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Dequantize(+multi-source accumulate): grid-stride threads over packs.
 // Sums d.nsrc compressed streams in T precision in stream order (matching the
@@ -641,11 +887,31 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       // 3x slower per element than the fp16 path's single int4)
       const int64_t full_subs = nq >> 2;
       const uint32_t B4 = oneb ? static_cast<uint32_t>(d.bucket >> 2) : 1u;
-      // four grid-stride iterations in flight: the decode chain is
-      // load(packed)+load(meta)->decode->store, so per-lane outstanding
-      // loads scale with the unroll depth (round-1 profile showed the
-      // 2-deep variant latency-bound at ~2.1 TB/s)
-      constexpr int U = 4;
+      if (oneb && small) {
+        // branch-free fast path (see deq_f32_fast)
+        const bool have = d.add != 0;
+        if (al16) {
+          if (have)
+            deq_f32_fast<BITS, true, true>(d, in0, meta_bytes, full_subs, B4,
+                                           t0, stride);
+          else
+            deq_f32_fast<BITS, true, false>(d, in0, meta_bytes, full_subs, B4,
+                                            t0, stride);
+        } else {
+          if (have)
+            deq_f32_fast<BITS, false, true>(d, in0, meta_bytes, full_subs, B4,
+                                            t0, stride);
+          else
+            deq_f32_fast<BITS, false, false>(d, in0, meta_bytes, full_subs,
+                                             B4, t0, stride);
+        }
+        goto f32_tail;
+      }
+      {
+      // two grid-stride iterations in flight (U=4 measured 0.170 ms vs
+      // 0.142 at U=2 for 64M/4-bit: the extra registers cost more occupancy
+      // than the added chains buy in latency hiding)
+      constexpr int U = 2;
       int64_t w = t0;
       for (; w + (U - 1) * stride < full_subs; w += U * stride) {
         int64_t ws2[U];
@@ -672,6 +938,10 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
           for (int u = 0; u < U; u++) {
             const int64_t g = ws2[u] >> 1;
             const int h = static_cast<int>(ws2[u] & 1);
+            // half-pack loads for even BITS (a full-pack load + register
+            // shift, which reads each pack twice, measured 0.156 ms vs
+            // 0.142 for 64M/4-bit: the doubled L1 traffic costs more than
+            // the narrower load saves)
             if constexpr ((BITS & 1) == 0) {
               value[u] = load_bytes(src + g * BITS + h * (BITS / 2), BITS / 2);
             } else {
@@ -744,6 +1014,8 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
         }
         store4f(outp, al16, v);
       }
+      }  // generic fp32 path
+    f32_tail:
       // scalar tail: elements [full_subs*4, nq)
       if ((nq & 3) && t0 == 0) {
         R* outr = reinterpret_cast<R*>(d.out);
@@ -776,6 +1048,26 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
         }
       }
       continue;  // next slice (16-bit loop below is for 2-byte dtypes)
+    }
+    if (oneb && small) {
+      // branch-free fast path (see deq_16_fast)
+      const bool have = d.add != 0;
+      if (al16) {
+        if (have)
+          deq_16_fast<T, BITS, true, true>(d, in0, meta_bytes, full_groups,
+                                           B8, t0, stride);
+        else
+          deq_16_fast<T, BITS, true, false>(d, in0, meta_bytes, full_groups,
+                                            B8, t0, stride);
+      } else {
+        if (have)
+          deq_16_fast<T, BITS, false, true>(d, in0, meta_bytes, full_groups,
+                                            B8, t0, stride);
+        else
+          deq_16_fast<T, BITS, false, false>(d, in0, meta_bytes, full_groups,
+                                             B8, t0, stride);
+      }
+      goto tail16;
     }
     for (int64_t g = t0; g < full_groups; g += stride) {
       uint32_t v[8];
@@ -813,6 +1105,7 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       store8<T>(outp, al16, v);
     }
 
+  tail16:
     // tail group (fewer than 8 elements): one thread, scalar
     const int mtail = static_cast<int>(nq - full_groups * 8);
     if (mtail > 0 && t0 == 0) {
@@ -931,9 +1224,21 @@ __global__ __launch_bounds__(kThreads) void k_add(const T* __restrict__ x,
   }
 }
 
+inline int max_blocks() {
+  // CGX_MAX_BLOCKS: experimentation knob for grid-size sweeps (default
+  // kMaxBlocks = 4096, i.e. 2 resident workgroup generations on 256 CUs)
+  static const int v = [] {
+    const char* e = std::getenv("CGX_MAX_BLOCKS");
+    const int x = e && *e ? std::atoi(e) : 0;
+    return x > 0 ? x : kMaxBlocks;
+  }();
+  return v;
+}
+
 inline int grid_for(int64_t work, int per_block) {
   const int64_t blocks = (work + per_block - 1) / per_block;
-  return static_cast<int>(blocks < 1 ? 1 : (blocks > kMaxBlocks ? kMaxBlocks : blocks));
+  const int64_t cap = max_blocks();
+  return static_cast<int>(blocks < 1 ? 1 : (blocks > cap ? cap : blocks));
 }
 
 #define CGX_DISPATCH_BITS(BITS_VAL, ...)                        \
