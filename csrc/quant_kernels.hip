@@ -313,6 +313,98 @@ __device__ __forceinline__ uint64_t load_bytes(const uint8_t* p, int nb) {
   return v;
 }
 
+// Register-resident quantize of one FULL bucket (bucket % 8 == 0, 16B
+// aligned, no error feedback) with a COMPILE-TIME groups-per-lane count G:
+// the runtime-trip stash loops of the original fast path compiled into
+// s_set_gpr_idx register-indexed access and exec-mask churn; with G known
+// the stash is plain unrolled registers.
+template <typename T, int BITS, int G>
+__device__ __forceinline__ void quant_bucket_reg(
+    const T* __restrict__ in, int lane, int ngroups, int64_t lb,
+    int64_t nb_slice, uint8_t* __restrict__ out, int64_t gbase,
+    float divisor, uint64_t seed, int stochastic, int slice_idx) {
+  using R = typename RawOf<T>::type;
+  uint32_t stash[G][8];
+  float lmin = INFINITY, lmax = -INFINITY;
+  if constexpr (sizeof(T) == 2) {
+    using PK = Pk2<T>;
+    typename PK::P pmin = bitcast<typename PK::P>(PK::kInf);
+    typename PK::P pmax = bitcast<typename PK::P>(PK::kNInf);
+#pragma unroll
+    for (int k = 0; k < G; k++) {
+      const int g = lane + k * kWave;
+      if (g >= ngroups) continue;  // buckets < 512: lanes beyond ngroups idle
+      const int4 a = *reinterpret_cast<const int4*>(
+          __builtin_assume_aligned(in + g * 8, 16));
+      const uint32_t w[4] = {static_cast<uint32_t>(a.x),
+                             static_cast<uint32_t>(a.y),
+                             static_cast<uint32_t>(a.z),
+                             static_cast<uint32_t>(a.w)};
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        const auto v = bitcast<typename PK::P>(w[q]);
+        pmin = PK::min(pmin, v);
+        pmax = PK::max(pmax, v);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        stash[k][j] = (w[j >> 1] >> ((j & 1) * 16)) & 0xFFFF;
+    }
+    lmin = fminf(PK::lo(pmin), PK::hi(pmin));
+    lmax = fmaxf(PK::lo(pmax), PK::hi(pmax));
+  } else {
+#pragma unroll
+    for (int k = 0; k < G; k++) {
+      const int g = lane + k * kWave;
+      if (g >= ngroups) continue;  // buckets < 512: lanes beyond ngroups idle
+      load8<T>(in + g * 8, true, stash[k]);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float f = raw2f<T>(stash[k][j]);
+        lmin = fminf(lmin, f);
+        lmax = fmaxf(lmax, f);
+      }
+    }
+  }
+  wave_minmax(lmin, lmax);
+  const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
+  const float unitf = raw2f<T>(unit_raw);
+  const float minf = lmin;
+  R* meta = reinterpret_cast<R*>(out);
+  if (lane == 0) {
+    meta[2 * lb] = static_cast<R>(unit_raw);
+    meta[2 * lb + 1] = static_cast<R>(f2raw<T>(lmin));
+  }
+  uint8_t* packed = out + 2 * sizeof(R) * nb_slice;
+  const bool live = unitf >= kEps;
+  const float rinv = live ? 1.0f / unitf : 0.0f;
+  using acc_t =
+      typename std::conditional<(BITS <= 4), uint32_t, uint64_t>::type;
+#pragma unroll
+  for (int k = 0; k < G; k++) {
+    const int g = lane + k * kWave;
+    if (g >= ngroups) continue;  // buckets < 512: lanes beyond ngroups idle
+    acc_t value = 0;
+    if (live) {
+      const uint64_t pr =
+          stochastic
+              ? rand_pack(seed, (static_cast<uint64_t>(slice_idx) << 44) |
+                                    static_cast<uint64_t>(gbase + g))
+              : 0;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
+        const float dd = (raw2f<T>(stash[k][j]) - minf) * rinv + rnd;
+        const uint32_t level =
+            static_cast<uint32_t>(fminf(floorf(dd), divisor));
+        value |= static_cast<acc_t>(level & ((1u << BITS) - 1)) << (j * BITS);
+      }
+    }
+    store_bytes(packed + (gbase + g) * BITS, static_cast<uint64_t>(value),
+                BITS);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Quantize: one wave per bucket.  ENCODE=true requires bucket % 8 == 0 for
 // every slice (host guarantees); ENCODE=false computes meta only (the
@@ -353,6 +445,33 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     // fast path: full bucket, whole groups per lane (wave-uniform branch)
     const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
     T* const fbp = d.fb ? reinterpret_cast<T*>(d.fb) + bstart : nullptr;
+    if (ENCODE && full && ngroups <= MAXG * kWave && !fbp) {
+      // fully-unrolled register path, dispatched on groups-per-lane
+      const int64_t gbase = bstart >> 3;
+      switch ((ngroups + kWave - 1) >> 6) {
+        case 1:
+          quant_bucket_reg<T, BITS, 1>(in, lane, ngroups, lb, nb_slice,
+                                       reinterpret_cast<uint8_t*>(d.out),
+                                       gbase, divisor, seed, stochastic, lo);
+          break;
+        case 2:
+          quant_bucket_reg<T, BITS, 2>(in, lane, ngroups, lb, nb_slice,
+                                       reinterpret_cast<uint8_t*>(d.out),
+                                       gbase, divisor, seed, stochastic, lo);
+          break;
+        case 3:
+          quant_bucket_reg<T, BITS, 3>(in, lane, ngroups, lb, nb_slice,
+                                       reinterpret_cast<uint8_t*>(d.out),
+                                       gbase, divisor, seed, stochastic, lo);
+          break;
+        default:
+          quant_bucket_reg<T, BITS, 4>(in, lane, ngroups, lb, nb_slice,
+                                       reinterpret_cast<uint8_t*>(d.out),
+                                       gbase, divisor, seed, stochastic, lo);
+          break;
+      }
+      continue;
+    }
     if (full && ngroups <= MAXG * kWave && !fbp) {
       if constexpr (sizeof(T) == 2) {
         using PK = Pk2<T>;
