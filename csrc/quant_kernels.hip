@@ -313,97 +313,147 @@ __device__ __forceinline__ uint64_t load_bytes(const uint8_t* p, int nb) {
   return v;
 }
 
-// Register-resident quantize of one FULL bucket (bucket % 8 == 0, 16B
-// aligned, no error feedback) with a COMPILE-TIME groups-per-lane count G:
-// the runtime-trip stash loops of the original fast path compiled into
-// s_set_gpr_idx register-indexed access and exec-mask churn; with G known
-// the stash is plain unrolled registers.
+// Register-resident quantize of a RUN of full buckets (bucket % 8 == 0, 16B
+// aligned, no error feedback) with a COMPILE-TIME groups-per-lane count G.
+// Two design points vs the round-1 per-bucket fast path:
+//  * G is a template parameter, so the stash loops fully unroll (the
+//    runtime-trip form compiled into s_set_gpr_idx register-indexed access);
+//  * the wave pipelines bucket i+1's HBM loads under bucket i's
+//    reduce+encode (two stash banks, statically unrolled 2x steady state) —
+//    the PMC profile showed SQ_WAIT_ANY ~17x SQ_BUSY on the serial
+//    load->reduce->encode chain.
+// 16-bit dtypes stash the RAW dwords (w[G][4], half the registers) and run
+// packed v_pk_min/max on them; elements are extracted at encode time.
 template <typename T, int BITS, int G>
-__device__ __forceinline__ void quant_bucket_reg(
-    const T* __restrict__ in, int lane, int ngroups, int64_t lb,
-    int64_t nb_slice, uint8_t* __restrict__ out, int64_t gbase,
-    float divisor, uint64_t seed, int stochastic, int slice_idx) {
+struct QuantRun {
   using R = typename RawOf<T>::type;
-  uint32_t stash[G][8];
-  float lmin = INFINITY, lmax = -INFINITY;
-  if constexpr (sizeof(T) == 2) {
-    using PK = Pk2<T>;
-    typename PK::P pmin = bitcast<typename PK::P>(PK::kInf);
-    typename PK::P pmax = bitcast<typename PK::P>(PK::kNInf);
+  static constexpr int kWords = sizeof(T) == 4 ? 8 : 4;
+  struct Stash {
+    uint32_t w[G][kWords];
+  };
+
+  const T* __restrict__ base;
+  R* __restrict__ meta;
+  uint8_t* __restrict__ packed;
+  int lane;
+  int ngroups;  // per bucket
+  float divisor;
+  uint64_t seed;
+  int stochastic;
+  int slice_idx;
+
+  __device__ __forceinline__ void load(Stash& s, int64_t lb) const {
+    const T* in = base + lb * (int64_t)(ngroups * 8);
 #pragma unroll
     for (int k = 0; k < G; k++) {
       const int g = lane + k * kWave;
-      if (g >= ngroups) continue;  // buckets < 512: lanes beyond ngroups idle
+      if (g >= ngroups) continue;  // buckets < 512: extra lanes idle
       const int4 a = *reinterpret_cast<const int4*>(
           __builtin_assume_aligned(in + g * 8, 16));
-      const uint32_t w[4] = {static_cast<uint32_t>(a.x),
-                             static_cast<uint32_t>(a.y),
-                             static_cast<uint32_t>(a.z),
-                             static_cast<uint32_t>(a.w)};
-#pragma unroll
-      for (int q = 0; q < 4; q++) {
-        const auto v = bitcast<typename PK::P>(w[q]);
-        pmin = PK::min(pmin, v);
-        pmax = PK::max(pmax, v);
+      if constexpr (sizeof(T) == 4) {
+        const int4 b = *reinterpret_cast<const int4*>(
+            __builtin_assume_aligned(in + g * 8 + 4, 16));
+        s.w[k][0] = a.x; s.w[k][1] = a.y; s.w[k][2] = a.z; s.w[k][3] = a.w;
+        s.w[k][4] = b.x; s.w[k][5] = b.y; s.w[k][6] = b.z; s.w[k][7] = b.w;
+      } else {
+        s.w[k][0] = a.x; s.w[k][1] = a.y; s.w[k][2] = a.z; s.w[k][3] = a.w;
       }
-#pragma unroll
-      for (int j = 0; j < 8; j++)
-        stash[k][j] = (w[j >> 1] >> ((j & 1) * 16)) & 0xFFFF;
     }
-    lmin = fminf(PK::lo(pmin), PK::hi(pmin));
-    lmax = fmaxf(PK::lo(pmax), PK::hi(pmax));
-  } else {
+  }
+
+  __device__ __forceinline__ uint32_t elem(const Stash& s, int k,
+                                           int j) const {
+    if constexpr (sizeof(T) == 4) {
+      return s.w[k][j];
+    } else {
+      return (s.w[k][j >> 1] >> ((j & 1) * 16)) & 0xFFFF;
+    }
+  }
+
+  __device__ __forceinline__ void encode(const Stash& s, int64_t lb) const {
+    float lmin = INFINITY, lmax = -INFINITY;
+    if constexpr (sizeof(T) == 2) {
+      using PK = Pk2<T>;
+      typename PK::P pmin = bitcast<typename PK::P>(PK::kInf);
+      typename PK::P pmax = bitcast<typename PK::P>(PK::kNInf);
+#pragma unroll
+      for (int k = 0; k < G; k++) {
+        const int g = lane + k * kWave;
+        if (g >= ngroups) continue;
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+          const auto v = bitcast<typename PK::P>(s.w[k][q]);
+          pmin = PK::min(pmin, v);
+          pmax = PK::max(pmax, v);
+        }
+      }
+      lmin = fminf(PK::lo(pmin), PK::hi(pmin));
+      lmax = fmaxf(PK::lo(pmax), PK::hi(pmax));
+    } else {
+#pragma unroll
+      for (int k = 0; k < G; k++) {
+        const int g = lane + k * kWave;
+        if (g >= ngroups) continue;
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          const float f = raw2f<T>(s.w[k][j]);
+          lmin = fminf(lmin, f);
+          lmax = fmaxf(lmax, f);
+        }
+      }
+    }
+    wave_minmax(lmin, lmax);
+    const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
+    const float unitf = raw2f<T>(unit_raw);
+    const float minf = lmin;
+    if (lane == 0) {
+      meta[2 * lb] = static_cast<R>(unit_raw);
+      meta[2 * lb + 1] = static_cast<R>(f2raw<T>(lmin));
+    }
+    const bool live = unitf >= kEps;
+    const float rinv = live ? 1.0f / unitf : 0.0f;
+    const int64_t gbase = lb * (int64_t)ngroups;
+    using acc_t =
+        typename std::conditional<(BITS <= 4), uint32_t, uint64_t>::type;
 #pragma unroll
     for (int k = 0; k < G; k++) {
       const int g = lane + k * kWave;
-      if (g >= ngroups) continue;  // buckets < 512: lanes beyond ngroups idle
-      load8<T>(in + g * 8, true, stash[k]);
+      if (g >= ngroups) continue;
+      acc_t value = 0;
+      if (live) {
+        const uint64_t pr =
+            stochastic
+                ? rand_pack(seed, (static_cast<uint64_t>(slice_idx) << 44) |
+                                      static_cast<uint64_t>(gbase + g))
+                : 0;
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        const float f = raw2f<T>(stash[k][j]);
-        lmin = fminf(lmin, f);
-        lmax = fmaxf(lmax, f);
+        for (int j = 0; j < 8; j++) {
+          const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
+          const float dd = (raw2f<T>(elem(s, k, j)) - minf) * rinv + rnd;
+          const uint32_t level =
+              static_cast<uint32_t>(fminf(floorf(dd), divisor));
+          value |=
+              static_cast<acc_t>(level & ((1u << BITS) - 1)) << (j * BITS);
+        }
       }
+      store_bytes(packed + (gbase + g) * BITS, static_cast<uint64_t>(value),
+                  BITS);
     }
   }
-  wave_minmax(lmin, lmax);
-  const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
-  const float unitf = raw2f<T>(unit_raw);
-  const float minf = lmin;
-  R* meta = reinterpret_cast<R*>(out);
-  if (lane == 0) {
-    meta[2 * lb] = static_cast<R>(unit_raw);
-    meta[2 * lb + 1] = static_cast<R>(f2raw<T>(lmin));
-  }
-  uint8_t* packed = out + 2 * sizeof(R) * nb_slice;
-  const bool live = unitf >= kEps;
-  const float rinv = live ? 1.0f / unitf : 0.0f;
-  using acc_t =
-      typename std::conditional<(BITS <= 4), uint32_t, uint64_t>::type;
-#pragma unroll
-  for (int k = 0; k < G; k++) {
-    const int g = lane + k * kWave;
-    if (g >= ngroups) continue;  // buckets < 512: lanes beyond ngroups idle
-    acc_t value = 0;
-    if (live) {
-      const uint64_t pr =
-          stochastic
-              ? rand_pack(seed, (static_cast<uint64_t>(slice_idx) << 44) |
-                                    static_cast<uint64_t>(gbase + g))
-              : 0;
-#pragma unroll
-      for (int j = 0; j < 8; j++) {
-        const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
-        const float dd = (raw2f<T>(stash[k][j]) - minf) * rinv + rnd;
-        const uint32_t level =
-            static_cast<uint32_t>(fminf(floorf(dd), divisor));
-        value |= static_cast<acc_t>(level & ((1u << BITS) - 1)) << (j * BITS);
-      }
+
+  // count buckets at lb0, lb0+nw, ...  Single stash bank: a two-bank
+  // software pipeline (load i+1 under encode i) measured fp32 0.098->0.123
+  // ms at 64M/4-bit — the doubled stash registers cost more occupancy than
+  // the prefetch hides, the same trade seen on the dequant ILP experiment.
+  __device__ __forceinline__ void run(int64_t lb0, int64_t count,
+                                      int64_t nw) const {
+    Stash A;
+    for (int64_t i = 0; i < count; i++) {
+      load(A, lb0 + i * nw);
+      encode(A, lb0 + i * nw);
     }
-    store_bytes(packed + (gbase + g) * BITS, static_cast<uint64_t>(value),
-                BITS);
   }
-}
+};
 
 // ---------------------------------------------------------------------------
 // Quantize: one wave per bucket.  ENCODE=true requires bucket % 8 == 0 for
@@ -446,30 +496,28 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
     T* const fbp = d.fb ? reinterpret_cast<T*>(d.fb) + bstart : nullptr;
     if (ENCODE && full && ngroups <= MAXG * kWave && !fbp) {
-      // fully-unrolled register path, dispatched on groups-per-lane
-      const int64_t gbase = bstart >> 3;
+      // pipelined register path over every eligible bucket this wave owns
+      // in this slice (see QuantRun)
+      const int64_t nb_full = nq / d.bucket;  // partial tail -> generic path
+      const int64_t count = (nb_full - lb + nw - 1) / nw;
+#define CGX_QRUN(GV)                                                       \
+  do {                                                                     \
+    QuantRun<T, BITS, GV> qr{reinterpret_cast<const T*>(d.in),             \
+                             reinterpret_cast<R*>(d.out),                  \
+                             reinterpret_cast<uint8_t*>(d.out) +           \
+                                 2 * sizeof(R) * nb_slice,                 \
+                             lane,    ngroups, divisor,                    \
+                             seed,    stochastic, lo};                     \
+    qr.run(lb, count, nw);                                                 \
+  } while (0)
       switch ((ngroups + kWave - 1) >> 6) {
-        case 1:
-          quant_bucket_reg<T, BITS, 1>(in, lane, ngroups, lb, nb_slice,
-                                       reinterpret_cast<uint8_t*>(d.out),
-                                       gbase, divisor, seed, stochastic, lo);
-          break;
-        case 2:
-          quant_bucket_reg<T, BITS, 2>(in, lane, ngroups, lb, nb_slice,
-                                       reinterpret_cast<uint8_t*>(d.out),
-                                       gbase, divisor, seed, stochastic, lo);
-          break;
-        case 3:
-          quant_bucket_reg<T, BITS, 3>(in, lane, ngroups, lb, nb_slice,
-                                       reinterpret_cast<uint8_t*>(d.out),
-                                       gbase, divisor, seed, stochastic, lo);
-          break;
-        default:
-          quant_bucket_reg<T, BITS, 4>(in, lane, ngroups, lb, nb_slice,
-                                       reinterpret_cast<uint8_t*>(d.out),
-                                       gbase, divisor, seed, stochastic, lo);
-          break;
+        case 1: CGX_QRUN(1); break;
+        case 2: CGX_QRUN(2); break;
+        case 3: CGX_QRUN(3); break;
+        default: CGX_QRUN(4); break;
       }
+#undef CGX_QRUN
+      b += (count - 1) * nw;  // outer loop adds one more nw
       continue;
     }
     if (full && ngroups <= MAXG * kWave && !fbp) {

@@ -123,8 +123,9 @@ def test_stochastic_rounding_statistics():
 
 
 def test_quantize_throughput_floor():
-    """Catastrophic-regression guard: 64M fp32 4-bit quantize must stream at
-    >= 1 TB/s effective input bandwidth on MI355X (roofline ~6 TB/s)."""
+    """Regression guard: 64M fp32 4-bit quantize must stream at >= 1.8 TB/s
+    effective input bandwidth on MI355X (round-2 measured 2.4-2.6 TB/s;
+    write-only roofline ~6.5 TB/s)."""
     from torch_cgx_amd import _C
     n, bits, bucket = 64 << 20, 4, 1024
     x = torch.randn(n, device=_dev())
@@ -140,7 +141,7 @@ def test_quantize_throughput_floor():
     dt = (time.perf_counter() - t0) / reps
     gbps = n * 4 / dt / 1e9
     print(f"quantize 64M fp32 4-bit: {dt*1e3:.2f} ms, {gbps:.0f} GB/s input")
-    assert gbps > 1000, f"quantize too slow: {gbps:.0f} GB/s"
+    assert gbps > 1800, f"quantize too slow: {gbps:.0f} GB/s"
 
 
 def test_dequantize_throughput_floor():
@@ -161,7 +162,8 @@ def test_dequantize_throughput_floor():
     dt = (time.perf_counter() - t0) / reps
     gbps = n * 4 / dt / 1e9
     print(f"dequantize 64M fp32 4-bit: {dt*1e3:.2f} ms, {gbps:.0f} GB/s out")
-    assert gbps > 1000, f"dequantize too slow: {gbps:.0f} GB/s"
+    # round-2 branch-free fast path measured ~3.0 TB/s output rate
+    assert gbps > 2400, f"dequantize too slow: {gbps:.0f} GB/s"
 
 
 @pytest.mark.parametrize("dtype", DTYPES)
