@@ -243,12 +243,45 @@ Engine::Engine(int rank, int size, bool is_cross)
 }
 
 Engine::~Engine() {
+  timer_drain();
   for (auto& s : slots_)
     if (s.done_ev) (void)hipEventDestroy(s.done_ev);
   for (auto& e : evs_)
     if (e) (void)hipEventDestroy(e);
   if (comm_stream_) (void)hipStreamDestroy(comm_stream_);
   if (deq_stream_) (void)hipStreamDestroy(deq_stream_);
+  if (timer_.inited) {
+    for (int r = 0; r < PhaseTimer::kRing; r++)
+      for (int e = 0; e < PhaseTimer::kEv; e++)
+        if (timer_.ev[r][e]) (void)hipEventDestroy(timer_.ev[r][e]);
+  }
+}
+
+void Engine::timer_drain() {
+  // collect any in-flight timing slots and print the final averages (short
+  // runs would otherwise never reach the 50-chunk periodic print)
+  PhaseTimer& t = timer_;
+  if (!t.inited) return;
+  for (int r = 0; r < PhaseTimer::kRing; r++) {
+    if (!t.pending[r]) continue;
+    if (hipEventSynchronize(t.ev[r][PhaseTimer::kEv - 1]) != hipSuccess)
+      continue;
+    for (int p = 0; p + 1 < PhaseTimer::kEv; p++) {
+      float ms = 0.f;
+      if (hipEventElapsedTime(&ms, t.ev[r][p], t.ev[r][p + 1]) == hipSuccess)
+        t.sum_ms[p] += ms;
+    }
+    t.count++;
+    t.pending[r] = false;
+  }
+  if (t.count > 0) {
+    fprintf(stderr,
+            "[cgx timings rank %d, %lld chunks total] quantize %.3f ms | "
+            "comm1 %.3f | decode+requant %.3f | comm2 %.3f | decode2 %.3f\n",
+            rank_, (long long)t.count, t.sum_ms[0] / t.count,
+            t.sum_ms[1] / t.count, t.sum_ms[2] / t.count,
+            t.sum_ms[3] / t.count, t.sum_ms[4] / t.count);
+  }
 }
 
 hipEvent_t Engine::next_ev() {

@@ -268,6 +268,7 @@ class Engine {
   void timer_begin(hipStream_t qs, bool enabled);
   void timer_mark(int idx, hipStream_t stream);  // idx 1..5
   void timer_finish();
+  void timer_drain();  // destructor: flush in-flight slots + final print
   bool timing_ = false;
   int timer_slot_ = -1;
 
