@@ -130,17 +130,54 @@ at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
     QuantDesc d;
     int64_t cum[2];
   } hb;
+  auto stream =
+      c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.device().index());
+  auto upload = [&](const Blob& blob) {
+    return at::from_blob(const_cast<Blob*>(&blob), {(int64_t)sizeof(Blob)},
+                         at::TensorOptions().dtype(at::kByte))
+        .to(x.device());
+  };
+  const int32_t base_flags = skip_incomplete ? kFlagSkipIncomplete : 0;
+  const bool aligned =
+      (reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0;
+  // mirror the engine's launch split (run_quantize): lean fast kernel for
+  // the full buckets, generic kernel for the partial tail
+  if (bucket_size % 8 == 0 && !fbp && aligned && n >= bucket_size &&
+      bucket_size <= 2048) {  // register-stash limit (4 groups/lane)
+    hb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), nullptr, n,
+                     (int32_t)bucket_size, base_flags};
+    hb.cum[0] = 0;
+    hb.cum[1] = n / bucket_size;
+    auto dev = upload(hb);
+    const char* devp = static_cast<const char*>(dev.data_ptr());
+    launch_quantize_fast(
+        reinterpret_cast<const QuantDesc*>(devp),
+        reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
+        hb.cum[1], dt, (int)bits, (uint64_t)seed, stochastic, stream.stream(),
+        skip_incomplete && (n % bucket_size) != 0,
+        (int)((bucket_size / 8 + 63) / 64));
+    if (!skip_incomplete && (n % bucket_size) != 0) {
+      Blob tb;
+      tb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), nullptr, n,
+                       (int32_t)bucket_size, base_flags | kFlagTailOnly};
+      tb.cum[0] = 0;
+      tb.cum[1] = 1;
+      auto tdev = upload(tb);
+      const char* tdevp = static_cast<const char*>(tdev.data_ptr());
+      launch_quantize_batch(
+          reinterpret_cast<const QuantDesc*>(tdevp),
+          reinterpret_cast<const int64_t*>(tdevp + offsetof(Blob, cum)), 1, 1,
+          dt, (int)bits, (uint64_t)seed, stochastic, stream.stream(),
+          /*buckets_mult8=*/true, false);
+    }
+    return out;
+  }
   hb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), fbp, n,
-                   (int32_t)bucket_size,
-                   skip_incomplete ? kFlagSkipIncomplete : 0};
+                   (int32_t)bucket_size, base_flags};
   hb.cum[0] = 0;
   hb.cum[1] = skip_incomplete ? n / bucket_size
                               : (n + bucket_size - 1) / bucket_size;
-  auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
-                           at::TensorOptions().dtype(at::kByte))
-                 .to(x.device());
-  auto stream =
-      c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(x.device().index());
+  auto dev = upload(hb);
   const char* devp = static_cast<const char*>(dev.data_ptr());
   launch_quantize_batch(
       reinterpret_cast<const QuantDesc*>(devp),

@@ -38,6 +38,9 @@ inline int64_t buffer_size(int64_t n, DType dt, int bits, int bucket_size,
 }
 
 constexpr int32_t kFlagSkipIncomplete = 1;
+// The slice's full buckets were already quantized by the lean fast kernel;
+// this (generic-kernel) desc covers only the trailing partial bucket.
+constexpr int32_t kFlagTailOnly = 2;
 
 // One quantize work item: compress `n` elems at `in` into `out` bytes.
 // fb (optional): error-feedback residual of the same length/dtype as `in`;
@@ -81,6 +84,19 @@ void launch_quantize_batch(const QuantDesc* descs, const int64_t* cum,
                            int bits, uint64_t seed, bool stochastic,
                            hipStream_t stream, bool buckets_mult8,
                            bool any_residual = false);
+
+// Lean fast-path kernel: every slice has bucket % 8 == 0, a 16B-aligned
+// input base, no error feedback; cum counts FULL buckets per slice (the
+// partial tail travels through launch_quantize_batch with kFlagTailOnly).
+// Separate launch so its register footprint (and thus occupancy) is not
+// inflated by the generic/EF code paths.
+// max_gpl: max over slices of ceil((bucket/8)/64) — <=2 selects the
+// half-stash kernel variant (higher occupancy for buckets <= 1024).
+void launch_quantize_fast(const QuantDesc* descs, const int64_t* cum,
+                          int nslices, int64_t total_buckets, DType dt,
+                          int bits, uint64_t seed, bool stochastic,
+                          hipStream_t stream, bool any_residual = false,
+                          int max_gpl = 4);
 
 void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
                              int nslices, int64_t total_groups, DType dt,

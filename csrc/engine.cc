@@ -408,13 +408,44 @@ at::Tensor& Engine::feedback_buf(int64_t key, int phase, int64_t numel,
 void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
                           DType dt, hipStream_t stream, bool stochastic,
                           char* fb_base, int64_t fb_rebase) {
-  // group by (bits, bucket%8==0)
-  std::map<std::pair<int, bool>, std::vector<const Slice*>> groups;
+  // Three launch kinds per bits value:
+  //   0 = lean fast kernel (bucket%8==0, 16B-aligned input, no error
+  //       feedback, >=1 full bucket) — low register footprint, cum counts
+  //       FULL buckets; a partial tail re-enters kind 1 with kFlagTailOnly
+  //   1 = generic fused kernel (EF, unaligned, tails)
+  //   2 = meta + pack two-pass (bucket%8 != 0)
+  struct Ent {
+    const Slice* s;
+    void* fb;
+    int32_t flags;
+  };
+  std::map<std::pair<int, int>, std::vector<Ent>> groups;
   for (const auto& s : slices) {
     if (s.n <= 0) continue;
-    groups[{s.bits, (s.bucket % 8) == 0}].push_back(&s);
+    void* fb =
+        fb_base ? fb_base + (s.fb_off - fb_rebase) * elem_size(dt) : nullptr;
+    TORCH_CHECK(!fb_base || s.fb_off >= fb_rebase,
+                "cgx: error-feedback slice offset below buffer base");
+    TORCH_CHECK(!fb || s.bucket % 8 == 0,
+                "cgx: CGX_ERROR_FEEDBACK requires bucket_size % 8 == 0");
+    const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
+    if ((s.bucket % 8) != 0) {
+      groups[{s.bits, 2}].push_back(Ent{&s, fb, flags});
+      continue;
+    }
+    const bool aligned = (reinterpret_cast<uintptr_t>(s.data) & 15) == 0;
+    // bucket <= 2048: whole bucket fits the register stash (4 groups/lane)
+    if (!fb && aligned && s.n >= s.bucket && s.bucket <= 2048) {
+      groups[{s.bits, 0}].push_back(Ent{&s, nullptr, flags});
+      if (!s.skip_incomplete && (s.n % s.bucket) != 0)
+        groups[{s.bits, 1}].push_back(Ent{&s, nullptr,
+                                          flags | kFlagTailOnly});
+    } else {
+      groups[{s.bits, 1}].push_back(Ent{&s, fb, flags});
+    }
   }
   for (auto& [key, list] : groups) {
+    const auto [bits, kind] = key;
     const int nsl = (int)list.size();
     const size_t desc_bytes = sizeof(QuantDesc) * nsl;
     const size_t cum_bytes = sizeof(int64_t) * (nsl + 1);
@@ -424,28 +455,38 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
     cum[0] = 0;
     bool any_residual = false;
     for (int i = 0; i < nsl; i++) {
-      const Slice& s = *list[i];
-      const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
-      void* fb =
-          fb_base ? fb_base + (s.fb_off - fb_rebase) * elem_size(dt) : nullptr;
-      TORCH_CHECK(!fb_base || s.fb_off >= fb_rebase,
-                  "cgx: error-feedback slice offset below buffer base");
-      TORCH_CHECK(!fb || s.bucket % 8 == 0,
-                  "cgx: CGX_ERROR_FEEDBACK requires bucket_size % 8 == 0");
-      qd[i] = QuantDesc{s.data, out_base + s.comp_off, fb, s.n, s.bucket,
-                        flags};
-      if (s.skip_incomplete) {
-        cum[i + 1] = cum[i] + s.n / s.bucket;
+      const Slice& s = *list[i].s;
+      qd[i] = QuantDesc{s.data, out_base + s.comp_off, list[i].fb, s.n,
+                        s.bucket, list[i].flags};
+      int64_t nb;
+      if (kind == 0) {
+        nb = s.n / s.bucket;  // full buckets only
+        any_residual |= s.skip_incomplete && (s.n % s.bucket) != 0;
+      } else if (list[i].flags & kFlagTailOnly) {
+        nb = 1;
+      } else if (s.skip_incomplete) {
+        nb = s.n / s.bucket;
         any_residual |= (s.n % s.bucket) != 0;
       } else {
-        cum[i + 1] = cum[i] + (s.n + s.bucket - 1) / s.bucket;
+        nb = (s.n + s.bucket - 1) / s.bucket;
       }
+      cum[i + 1] = cum[i] + nb;
     }
     char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
-    launch_quantize_batch(reinterpret_cast<QuantDesc*>(dev),
-                          reinterpret_cast<int64_t*>(dev + desc_bytes), nsl,
-                          cum[nsl], dt, key.first, seed_++, stochastic, stream,
-                          key.second, any_residual);
+    auto* ddesc = reinterpret_cast<QuantDesc*>(dev);
+    auto* dcum = reinterpret_cast<int64_t*>(dev + desc_bytes);
+    if (kind == 0) {
+      int max_gpl = 1;
+      for (const auto& e : list)
+        max_gpl = std::max(max_gpl,
+                           ((e.s->bucket >> 3) + 63) >> 6);
+      launch_quantize_fast(ddesc, dcum, nsl, cum[nsl], dt, bits, seed_++,
+                           stochastic, stream, any_residual, max_gpl);
+    } else {
+      launch_quantize_batch(ddesc, dcum, nsl, cum[nsl], dt, bits, seed_++,
+                            stochastic, stream, /*buckets_mult8=*/kind == 1,
+                            any_residual);
+    }
   }
 }
 

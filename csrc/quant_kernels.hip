@@ -455,6 +455,63 @@ struct QuantRun {
   }
 };
 
+// Lean fast-path kernel (see compress.h launch_quantize_fast): only the
+// QuantRun register path is instantiated, so the register allocator is not
+// forced to the generic/EF paths' footprint (k_quantize<float,4,true>
+// allocates 139 VGPRs = 3 waves/SIMD; this kernel ~half that).
+template <typename T, int BITS, int MAXGT>
+__global__ __launch_bounds__(kThreads) void k_quantize_fast(
+    const QuantDesc* __restrict__ descs, const int64_t* __restrict__ cum,
+    int nslices, int64_t total_buckets, uint64_t seed, int stochastic) {
+  using R = typename RawOf<T>::type;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wid = static_cast<int64_t>(blockIdx.x) * (kThreads / kWave) +
+                      (threadIdx.x / kWave);
+  const int64_t nw = static_cast<int64_t>(gridDim.x) * (kThreads / kWave);
+  constexpr float divisor = static_cast<float>((1 << BITS) - 1);
+  int64_t b = wid;
+  while (b < total_buckets) {
+    int lo = 0, hi = nslices;
+    while (hi - lo > 1) {
+      const int mid = (lo + hi) >> 1;
+      if (cum[mid] <= b) lo = mid; else hi = mid;
+    }
+    const QuantDesc d = descs[lo];
+    const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
+    const int64_t nb_slice =
+        skip ? d.n / d.bucket
+             : (d.n + d.bucket - 1) / static_cast<int64_t>(d.bucket);
+    const int64_t nb_full = cum[lo + 1] - cum[lo];
+    const int64_t lb = b - cum[lo];
+    const int ngroups = d.bucket >> 3;
+    const int64_t count = (nb_full - lb + nw - 1) / nw;
+#define CGX_QRUNF(GV)                                                      \
+  do {                                                                     \
+    QuantRun<T, BITS, GV> qr{reinterpret_cast<const T*>(d.in),             \
+                             reinterpret_cast<R*>(d.out),                  \
+                             reinterpret_cast<uint8_t*>(d.out) +           \
+                                 2 * sizeof(R) * nb_slice,                 \
+                             lane,    ngroups, divisor,                    \
+                             seed,    stochastic, lo};                     \
+    qr.run(lb, count, nw);                                                 \
+  } while (0)
+    if constexpr (MAXGT <= 2) {
+      // all buckets <= 1024: half the stash registers -> higher occupancy
+      if (((ngroups + kWave - 1) >> 6) <= 1) CGX_QRUNF(1);
+      else CGX_QRUNF(2);
+    } else {
+      switch ((ngroups + kWave - 1) >> 6) {
+        case 1: CGX_QRUNF(1); break;
+        case 2: CGX_QRUNF(2); break;
+        case 3: CGX_QRUNF(3); break;
+        default: CGX_QRUNF(4); break;
+      }
+    }
+#undef CGX_QRUNF
+    b += count * nw;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Quantize: one wave per bucket.  ENCODE=true requires bucket % 8 == 0 for
 // every slice (host guarantees); ENCODE=false computes meta only (the
@@ -482,8 +539,14 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
     const int64_t nq =
         skip ? (d.n / d.bucket) * static_cast<int64_t>(d.bucket) : d.n;
-    const int64_t nb_slice = cum[lo + 1] - cum[lo];
-    const int64_t lb = b - cum[lo];
+    int64_t nb_slice = cum[lo + 1] - cum[lo];
+    int64_t lb = b - cum[lo];
+    if (d.flags & kFlagTailOnly) {
+      // full buckets were handled by k_quantize_fast; this desc's single
+      // cum entry maps to the trailing partial bucket
+      lb += d.n / d.bucket;
+      nb_slice = (d.n + d.bucket - 1) / static_cast<int64_t>(d.bucket);
+    }
     const int64_t bstart = lb * static_cast<int64_t>(d.bucket);
     const int cur = static_cast<int>(min(static_cast<int64_t>(d.bucket), nq - bstart));
     const T* in = reinterpret_cast<const T*>(d.in) + bstart;
@@ -1453,6 +1516,33 @@ void launch_quantize_batch(const QuantDesc* descs, const int64_t* cum,
         hipLaunchKernelGGL((k_pack_generic<T, BITS>), dim3(grid),
                            dim3(kThreads), 0, stream, descs, nslices, seed,
                            (int)stochastic);
+      }
+    }
+    if (any_residual) {
+      hipLaunchKernelGGL((k_residual_q<T>), dim3(32), dim3(kThreads), 0,
+                         stream, descs, nslices, bits);
+    }
+  }));
+}
+
+void launch_quantize_fast(const QuantDesc* descs, const int64_t* cum,
+                          int nslices, int64_t total_buckets, DType dt,
+                          int bits, uint64_t seed, bool stochastic,
+                          hipStream_t stream, bool any_residual,
+                          int max_gpl) {
+  if (nslices <= 0) return;
+  const int grid = grid_for(std::max<int64_t>(total_buckets, 1),
+                            kThreads / kWave);
+  CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
+    if (total_buckets > 0) {
+      if (max_gpl <= 2) {
+        hipLaunchKernelGGL((k_quantize_fast<T, BITS, 2>), dim3(grid),
+                           dim3(kThreads), 0, stream, descs, cum, nslices,
+                           total_buckets, seed, (int)stochastic);
+      } else {
+        hipLaunchKernelGGL((k_quantize_fast<T, BITS, 4>), dim3(grid),
+                           dim3(kThreads), 0, stream, descs, cum, nslices,
+                           total_buckets, seed, (int)stochastic);
       }
     }
     if (any_residual) {
