@@ -202,16 +202,41 @@ void py_dequantize(at::Tensor comp, at::Tensor out, int64_t bits,
     DequantDesc d;
     int64_t cum[2];
   } hb;
-  hb.d = DequantDesc{comp.data_ptr<uint8_t>(), out.data_ptr(), n, 0,
-                     (int32_t)bucket_size, 1, add ? 1 : 0,
-                     skip_incomplete ? kFlagSkipIncomplete : 0};
-  hb.cum[0] = 0;
-  hb.cum[1] = (nq + 7) / 8;
-  auto dev = at::from_blob(&hb, {(int64_t)sizeof(Blob)},
-                           at::TensorOptions().dtype(at::kByte))
-                 .to(out.device());
   auto stream =
       c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(out.device().index());
+  auto upload = [&](const Blob& blob) {
+    return at::from_blob(const_cast<Blob*>(&blob), {(int64_t)sizeof(Blob)},
+                         at::TensorOptions().dtype(at::kByte))
+        .to(out.device());
+  };
+  const int32_t base_flags = skip_incomplete ? kFlagSkipIncomplete : 0;
+  hb.d = DequantDesc{comp.data_ptr<uint8_t>(), out.data_ptr(), n, 0,
+                     (int32_t)bucket_size, 1, add ? 1 : 0, base_flags};
+  hb.cum[0] = 0;
+  hb.cum[1] = (nq + 7) / 8;
+  // mirror the engine's launch split (run_dequant)
+  if (bucket_size % 8 == 0 && nq < (int64_t(1) << 28)) {
+    auto dev = upload(hb);
+    const char* devp = static_cast<const char*>(dev.data_ptr());
+    launch_dequantize_fast(reinterpret_cast<const DequantDesc*>(devp), 1,
+                           hb.cum[1], dt, (int)bits, stream.stream(),
+                           skip_incomplete && (n % bucket_size) != 0);
+    const bool ragged =
+        elem_size(dt) == 4 ? (nq & 3) != 0 : (nq & 7) != 0;
+    if (ragged) {
+      Blob tb = hb;
+      tb.d.flags = base_flags | kFlagTailOnly;
+      tb.cum[1] = 1;
+      auto tdev = upload(tb);
+      const char* tdevp = static_cast<const char*>(tdev.data_ptr());
+      launch_dequantize_batch(
+          reinterpret_cast<const DequantDesc*>(tdevp),
+          reinterpret_cast<const int64_t*>(tdevp + offsetof(Blob, cum)), 1, 1,
+          dt, (int)bits, stream.stream(), false);
+    }
+    return;
+  }
+  auto dev = upload(hb);
   const char* devp = static_cast<const char*>(dev.data_ptr());
   launch_dequantize_batch(
       reinterpret_cast<const DequantDesc*>(devp),
@@ -244,6 +269,25 @@ void py_dequantize_multi(at::Tensor comp, at::Tensor out, int64_t bits,
   auto stream =
       c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(out.device().index());
   const char* devp = static_cast<const char*>(dev.data_ptr());
+  if (bucket_size % 8 == 0 && n < (int64_t(1) << 28)) {
+    launch_dequantize_fast(reinterpret_cast<const DequantDesc*>(devp), 1,
+                           hb.cum[1], dt, (int)bits, stream.stream());
+    const bool ragged = elem_size(dt) == 4 ? (n & 3) != 0 : (n & 7) != 0;
+    if (ragged) {
+      Blob tb = hb;
+      tb.d.flags = kFlagTailOnly;
+      tb.cum[1] = 1;
+      auto tdev = at::from_blob(&tb, {(int64_t)sizeof(Blob)},
+                                at::TensorOptions().dtype(at::kByte))
+                      .to(out.device());
+      const char* tdevp = static_cast<const char*>(tdev.data_ptr());
+      launch_dequantize_batch(
+          reinterpret_cast<const DequantDesc*>(tdevp),
+          reinterpret_cast<const int64_t*>(tdevp + offsetof(Blob, cum)), 1, 1,
+          dt, (int)bits, stream.stream());
+    }
+    return;
+  }
   launch_dequantize_batch(
       reinterpret_cast<const DequantDesc*>(devp),
       reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,

@@ -103,6 +103,13 @@ void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
                              int bits, hipStream_t stream,
                              bool any_residual = false);
 
+// Lean dequant kernel: every slice has bucket % 8 == 0 and fits u32
+// indexing (n < 2^28); ragged tails (n % 4 fp32 / n % 8 16-bit) re-enter
+// launch_dequantize_batch with kFlagTailOnly.
+void launch_dequantize_fast(const DequantDesc* descs, int nslices,
+                            int64_t total_groups, DType dt, int bits,
+                            hipStream_t stream, bool any_residual = false);
+
 // y[i] += x[i] elementwise (T precision), n elements.
 void launch_add(const void* x, void* y, int64_t n, DType dt,
                 hipStream_t stream);

@@ -493,12 +493,31 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
 void Engine::run_dequant(const std::vector<Slice>& slices,
                          const uint8_t* in_base, int64_t src_stride, int nsrc,
                          bool add, DType dt, hipStream_t stream) {
-  std::map<int, std::vector<const Slice*>> groups;
+  // kind 0: lean fast kernel (bucket%8==0, u32-indexable); a ragged tail
+  // re-enters kind 1 with kFlagTailOnly.  kind 1: generic kernel.
+  struct Ent {
+    const Slice* s;
+    int32_t flags;
+  };
+  const int es = elem_size(dt);
+  std::map<std::pair<int, int>, std::vector<Ent>> groups;
   for (const auto& s : slices) {
     if (s.n <= 0) continue;
-    groups[s.bits].push_back(&s);
+    const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
+    const int64_t nq =
+        s.skip_incomplete ? s.n / s.bucket * (int64_t)s.bucket : s.n;
+    const bool fast = (s.bucket % 8) == 0 && nq < (int64_t(1) << 28);
+    if (fast) {
+      groups[{s.bits, 0}].push_back(Ent{&s, flags});
+      const bool ragged = es == 4 ? (nq & 3) != 0 : (nq & 7) != 0;
+      if (ragged)
+        groups[{s.bits, 1}].push_back(Ent{&s, flags | kFlagTailOnly});
+    } else {
+      groups[{s.bits, 1}].push_back(Ent{&s, flags});
+    }
   }
-  for (auto& [bits, list] : groups) {
+  for (auto& [key, list] : groups) {
+    const auto [bits, kind] = key;
     const int nsl = (int)list.size();
     const size_t desc_bytes = sizeof(DequantDesc) * nsl;
     const size_t cum_bytes = sizeof(int64_t) * (nsl + 1);
@@ -508,22 +527,32 @@ void Engine::run_dequant(const std::vector<Slice>& slices,
     cum[0] = 0;
     bool any_residual = false;
     for (int i = 0; i < nsl; i++) {
-      const Slice& s = *list[i];
-      const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
+      const Slice& s = *list[i].s;
       dd[i] = DequantDesc{in_base + s.comp_off, s.data,    s.n, src_stride,
-                          s.bucket,             nsrc, add ? 1 : 0, flags};
-      if (s.skip_incomplete) {
+                          s.bucket,             nsrc, add ? 1 : 0,
+                          list[i].flags};
+      int64_t work;
+      if (list[i].flags & kFlagTailOnly) {
+        work = 1;
+      } else if (s.skip_incomplete) {
         const int64_t nq = s.n / s.bucket * (int64_t)s.bucket;
-        cum[i + 1] = cum[i] + (nq + 7) / 8;
+        work = (nq + 7) / 8;
         any_residual |= (s.n % s.bucket) != 0;
       } else {
-        cum[i + 1] = cum[i] + (s.n + 7) / 8;
+        work = (s.n + 7) / 8;
       }
+      cum[i + 1] = cum[i] + work;
     }
     char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
-    launch_dequantize_batch(reinterpret_cast<DequantDesc*>(dev),
-                            reinterpret_cast<int64_t*>(dev + desc_bytes), nsl,
-                            cum[nsl], dt, bits, stream, any_residual);
+    auto* ddesc = reinterpret_cast<DequantDesc*>(dev);
+    auto* dcum = reinterpret_cast<int64_t*>(dev + desc_bytes);
+    if (kind == 0) {
+      launch_dequantize_fast(ddesc, nsl, cum[nsl], dt, bits, stream,
+                             any_residual);
+    } else {
+      launch_dequantize_batch(ddesc, dcum, nsl, cum[nsl], dt, bits, stream,
+                              any_residual);
+    }
   }
 }
 

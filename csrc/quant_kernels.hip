@@ -1075,6 +1075,68 @@ __device__ void deq_16_fast(const DequantDesc& d,
   }
 }
 
+// Lean dequantize kernel: only the branch-free fast paths (bucket%8==0,
+// u32-indexable slices) are instantiated — the fused k_dequant allocates
+// 102 VGPRs (fp32) because the register allocator must also cover the
+// generic/tail paths; ragged tails re-enter k_dequant with kFlagTailOnly.
+template <typename T, int BITS>
+__global__ __launch_bounds__(kThreads) void k_dequant_fast(
+    const DequantDesc* __restrict__ descs, int nslices) {
+  using R = typename RawOf<T>::type;
+  const int64_t t0 =
+      static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int si = 0; si < nslices; si++) {
+    const DequantDesc d = descs[si];
+    const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
+    const int64_t nq =
+        skip ? (d.n / d.bucket) * static_cast<int64_t>(d.bucket) : d.n;
+    const int64_t nb_slice =
+        skip ? d.n / d.bucket : (d.n + d.bucket - 1) / d.bucket;
+    const int64_t meta_bytes = 2 * sizeof(R) * nb_slice;
+    const bool al16 = (reinterpret_cast<uintptr_t>(d.out) & 15) == 0;
+    const uint8_t* const in0 = d.in + meta_bytes;
+    const bool have = d.add != 0;
+    if constexpr (sizeof(T) == 4) {
+      const int64_t full_subs = nq >> 2;
+      const uint32_t B4 = static_cast<uint32_t>(d.bucket >> 2);
+      if (al16) {
+        if (have)
+          deq_f32_fast<BITS, true, true>(d, in0, meta_bytes, full_subs, B4,
+                                         t0, stride);
+        else
+          deq_f32_fast<BITS, true, false>(d, in0, meta_bytes, full_subs, B4,
+                                          t0, stride);
+      } else {
+        if (have)
+          deq_f32_fast<BITS, false, true>(d, in0, meta_bytes, full_subs, B4,
+                                          t0, stride);
+        else
+          deq_f32_fast<BITS, false, false>(d, in0, meta_bytes, full_subs, B4,
+                                           t0, stride);
+      }
+    } else {
+      const int64_t full_groups = nq >> 3;
+      const uint32_t B8 = static_cast<uint32_t>(d.bucket >> 3);
+      if (al16) {
+        if (have)
+          deq_16_fast<T, BITS, true, true>(d, in0, meta_bytes, full_groups,
+                                           B8, t0, stride);
+        else
+          deq_16_fast<T, BITS, true, false>(d, in0, meta_bytes, full_groups,
+                                            B8, t0, stride);
+      } else {
+        if (have)
+          deq_16_fast<T, BITS, false, true>(d, in0, meta_bytes, full_groups,
+                                            B8, t0, stride);
+        else
+          deq_16_fast<T, BITS, false, false>(d, in0, meta_bytes, full_groups,
+                                             B8, t0, stride);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Dequantize(+multi-source accumulate): grid-stride threads over packs.
 // Sums d.nsrc compressed streams in T precision in stream order (matching the
@@ -1117,26 +1179,8 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       // 3x slower per element than the fp16 path's single int4)
       const int64_t full_subs = nq >> 2;
       const uint32_t B4 = oneb ? static_cast<uint32_t>(d.bucket >> 2) : 1u;
-      if (oneb && small) {
-        // branch-free fast path (see deq_f32_fast)
-        const bool have = d.add != 0;
-        if (al16) {
-          if (have)
-            deq_f32_fast<BITS, true, true>(d, in0, meta_bytes, full_subs, B4,
-                                           t0, stride);
-          else
-            deq_f32_fast<BITS, true, false>(d, in0, meta_bytes, full_subs, B4,
-                                            t0, stride);
-        } else {
-          if (have)
-            deq_f32_fast<BITS, false, true>(d, in0, meta_bytes, full_subs, B4,
-                                            t0, stride);
-          else
-            deq_f32_fast<BITS, false, false>(d, in0, meta_bytes, full_subs,
-                                             B4, t0, stride);
-        }
-        goto f32_tail;
-      }
+      // kFlagTailOnly: k_dequant_fast already decoded the full 4-elem subs
+      if (d.flags & kFlagTailOnly) goto f32_tail;
       {
       // two grid-stride iterations in flight (U=4 measured 0.170 ms vs
       // 0.142 at U=2 for 64M/4-bit: the extra registers cost more occupancy
@@ -1279,26 +1323,8 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
       }
       continue;  // next slice (16-bit loop below is for 2-byte dtypes)
     }
-    if (oneb && small) {
-      // branch-free fast path (see deq_16_fast)
-      const bool have = d.add != 0;
-      if (al16) {
-        if (have)
-          deq_16_fast<T, BITS, true, true>(d, in0, meta_bytes, full_groups,
-                                           B8, t0, stride);
-        else
-          deq_16_fast<T, BITS, true, false>(d, in0, meta_bytes, full_groups,
-                                            B8, t0, stride);
-      } else {
-        if (have)
-          deq_16_fast<T, BITS, false, true>(d, in0, meta_bytes, full_groups,
-                                            B8, t0, stride);
-        else
-          deq_16_fast<T, BITS, false, false>(d, in0, meta_bytes, full_groups,
-                                             B8, t0, stride);
-      }
-      goto tail16;
-    }
+    // kFlagTailOnly: k_dequant_fast already decoded the full 8-elem groups
+    if (d.flags & kFlagTailOnly) goto tail16;
     for (int64_t g = t0; g < full_groups; g += stride) {
       uint32_t v[8];
       bool have = d.add != 0;
@@ -1562,6 +1588,23 @@ void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
     if (total_groups > 0) {
       hipLaunchKernelGGL((k_dequant<T, BITS>), dim3(grid), dim3(kThreads), 0,
                          stream, descs, cum, nslices, total_groups);
+    }
+    if (any_residual) {
+      hipLaunchKernelGGL((k_residual_d<T>), dim3(32), dim3(kThreads), 0,
+                         stream, descs, nslices, bits);
+    }
+  }));
+}
+
+void launch_dequantize_fast(const DequantDesc* descs, int nslices,
+                            int64_t total_groups, DType dt, int bits,
+                            hipStream_t stream, bool any_residual) {
+  if (nslices <= 0) return;
+  const int grid = grid_for(std::max<int64_t>(total_groups, 1), kThreads);
+  CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
+    if (total_groups > 0) {
+      hipLaunchKernelGGL((k_dequant_fast<T, BITS>), dim3(grid),
+                         dim3(kThreads), 0, stream, descs, nslices);
     }
     if (any_residual) {
       hipLaunchKernelGGL((k_residual_d<T>), dim3(32), dim3(kThreads), 0,
