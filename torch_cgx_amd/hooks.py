@@ -61,7 +61,9 @@ def _allreduce_fut(process_group: dist.ProcessGroup, tensor: torch.Tensor
     group_to_use = (process_group if process_group is not None
                     else dist.group.WORLD)
     # Divide first to avoid fp16 overflow; the backend computes SUM.
-    tensor.div_(group_to_use.size())
+    # (ws=1: division by 1 is the identity — skip the kernel launch)
+    if group_to_use.size() > 1:
+        tensor.div_(group_to_use.size())
     return (dist.all_reduce(tensor, group=group_to_use, async_op=True)
             .get_future()
             .then(lambda fut: fut.value()[0]))
