@@ -133,7 +133,7 @@ def main():
     if distributed:
         model = nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank] if use_cuda else None,
-            bucket_cap_mb=25)
+            bucket_cap_mb=25, gradient_as_bucket_view=True)
         if args.backend == "cgx":
             import torch_cgx_amd
             os.environ["CGX_COMPRESSION_QUANTIZATION_BITS"] = str(bits)
