@@ -441,10 +441,8 @@ struct QuantRun {
     }
   }
 
-  // count buckets at lb0, lb0+nw, ...  Single stash bank: a two-bank
-  // software pipeline (load i+1 under encode i) measured fp32 0.098->0.123
-  // ms at 64M/4-bit — the doubled stash registers cost more occupancy than
-  // the prefetch hides, the same trade seen on the dequant ILP experiment.
+  // count buckets at lb0, lb0+nw, ...  Single stash bank (used by the
+  // MAXGT=4 kernel variant, where a second bank would cost occupancy).
   __device__ __forceinline__ void run(int64_t lb0, int64_t count,
                                       int64_t nw) const {
     Stash A;
@@ -452,6 +450,24 @@ struct QuantRun {
       load(A, lb0 + i * nw);
       encode(A, lb0 + i * nw);
     }
+  }
+
+  // Two-bank software pipeline: bucket i+1's HBM loads issue before bucket
+  // i's reduce+encode, hiding the load latency of the serial per-wave
+  // bucket chain.  Only profitable in the LEAN kernel at MAXGT<=2 (a 2-bank
+  // stash on the old 139-VGPR fused kernel measured 0.098->0.123 ms).
+  __device__ __forceinline__ void run_pipelined(int64_t lb0, int64_t count,
+                                                int64_t nw) const {
+    Stash A, B;
+    load(A, lb0);
+    int64_t i = 0;
+    for (; i + 2 <= count; i += 2) {
+      load(B, lb0 + (i + 1) * nw);
+      encode(A, lb0 + i * nw);
+      if (i + 2 < count) load(A, lb0 + (i + 2) * nw);
+      encode(B, lb0 + (i + 1) * nw);
+    }
+    if (i < count) encode(A, lb0 + i * nw);
   }
 };
 
@@ -496,7 +512,10 @@ __global__ __launch_bounds__(kThreads) void k_quantize_fast(
     qr.run(lb, count, nw);                                                 \
   } while (0)
     if constexpr (MAXGT <= 2) {
-      // all buckets <= 1024: half the stash registers -> higher occupancy
+      // all buckets <= 1024: half the stash registers -> higher occupancy.
+      // (A two-bank pipelined run was retried here after the lean-kernel
+      // split: VGPR 53 -> 94, quantize 0.072 -> 0.077 ms — the second bank
+      // still costs more occupancy than its prefetch hides.  Rejected.)
       if (((ngroups + kWave - 1) >> 6) <= 1) CGX_QRUNF(1);
       else CGX_QRUNF(2);
     } else {
