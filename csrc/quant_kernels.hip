@@ -452,23 +452,6 @@ struct QuantRun {
     }
   }
 
-  // Two-bank software pipeline: bucket i+1's HBM loads issue before bucket
-  // i's reduce+encode, hiding the load latency of the serial per-wave
-  // bucket chain.  Only profitable in the LEAN kernel at MAXGT<=2 (a 2-bank
-  // stash on the old 139-VGPR fused kernel measured 0.098->0.123 ms).
-  __device__ __forceinline__ void run_pipelined(int64_t lb0, int64_t count,
-                                                int64_t nw) const {
-    Stash A, B;
-    load(A, lb0);
-    int64_t i = 0;
-    for (; i + 2 <= count; i += 2) {
-      load(B, lb0 + (i + 1) * nw);
-      encode(A, lb0 + i * nw);
-      if (i + 2 < count) load(A, lb0 + (i + 2) * nw);
-      encode(B, lb0 + (i + 1) * nw);
-    }
-    if (i < count) encode(A, lb0 + i * nw);
-  }
 };
 
 // Lean fast-path kernel (see compress.h launch_quantize_fast): only the
