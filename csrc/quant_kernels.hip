@@ -560,31 +560,6 @@ __global__ __launch_bounds__(kThreads) void k_quantize(
     // fast path: full bucket, whole groups per lane (wave-uniform branch)
     const bool full = cur == d.bucket && (cur & 7) == 0 && al16;
     T* const fbp = d.fb ? reinterpret_cast<T*>(d.fb) + bstart : nullptr;
-    if (ENCODE && full && ngroups <= MAXG * kWave && !fbp) {
-      // pipelined register path over every eligible bucket this wave owns
-      // in this slice (see QuantRun)
-      const int64_t nb_full = nq / d.bucket;  // partial tail -> generic path
-      const int64_t count = (nb_full - lb + nw - 1) / nw;
-#define CGX_QRUN(GV)                                                       \
-  do {                                                                     \
-    QuantRun<T, BITS, GV> qr{reinterpret_cast<const T*>(d.in),             \
-                             reinterpret_cast<R*>(d.out),                  \
-                             reinterpret_cast<uint8_t*>(d.out) +           \
-                                 2 * sizeof(R) * nb_slice,                 \
-                             lane,    ngroups, divisor,                    \
-                             seed,    stochastic, lo};                     \
-    qr.run(lb, count, nw);                                                 \
-  } while (0)
-      switch ((ngroups + kWave - 1) >> 6) {
-        case 1: CGX_QRUN(1); break;
-        case 2: CGX_QRUN(2); break;
-        case 3: CGX_QRUN(3); break;
-        default: CGX_QRUN(4); break;
-      }
-#undef CGX_QRUN
-      b += (count - 1) * nw;  // outer loop adds one more nw
-      continue;
-    }
     if (full && ngroups <= MAXG * kWave && !fbp) {
       if constexpr (sizeof(T) == 2) {
         using PK = Pk2<T>;
