@@ -150,12 +150,21 @@ at::Tensor py_quantize(at::Tensor x, int64_t bits, int64_t bucket_size,
     hb.cum[1] = n / bucket_size;
     auto dev = upload(hb);
     const char* devp = static_cast<const char*>(dev.data_ptr());
-    launch_quantize_fast(
-        reinterpret_cast<const QuantDesc*>(devp),
-        reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
-        hb.cum[1], dt, (int)bits, (uint64_t)seed, stochastic, stream.stream(),
-        skip_incomplete && (n % bucket_size) != 0,
-        (int)((bucket_size / 8 + 63) / 64));
+    const int64_t ng = bucket_size >> 3;
+    if (ng < 64 && (ng & (ng - 1)) == 0) {
+      launch_quantize_sub(
+          reinterpret_cast<const QuantDesc*>(devp),
+          reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
+          hb.cum[1], dt, (int)bits, (uint64_t)seed, stochastic,
+          stream.stream(), skip_incomplete && (n % bucket_size) != 0);
+    } else {
+      launch_quantize_fast(
+          reinterpret_cast<const QuantDesc*>(devp),
+          reinterpret_cast<const int64_t*>(devp + offsetof(Blob, cum)), 1,
+          hb.cum[1], dt, (int)bits, (uint64_t)seed, stochastic,
+          stream.stream(), skip_incomplete && (n % bucket_size) != 0,
+          (int)((bucket_size / 8 + 63) / 64));
+    }
     if (!skip_incomplete && (n % bucket_size) != 0) {
       Blob tb;
       tb.d = QuantDesc{x.data_ptr(), out.data_ptr<uint8_t>(), nullptr, n,

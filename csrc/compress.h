@@ -98,6 +98,14 @@ void launch_quantize_fast(const QuantDesc* descs, const int64_t* cum,
                           hipStream_t stream, bool any_residual = false,
                           int max_gpl = 4);
 
+// Small-bucket variant of the fast kernel: every slice has bucket/8 a
+// power of two < 64; the wave packs 64/(bucket/8) buckets per iteration
+// (segmented DPP reductions).  Separate kernel for register isolation.
+void launch_quantize_sub(const QuantDesc* descs, const int64_t* cum,
+                         int nslices, int64_t total_buckets, DType dt,
+                         int bits, uint64_t seed, bool stochastic,
+                         hipStream_t stream, bool any_residual = false);
+
 void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
                              int nslices, int64_t total_groups, DType dt,
                              int bits, hipStream_t stream,

@@ -434,9 +434,13 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
       continue;
     }
     const bool aligned = (reinterpret_cast<uintptr_t>(s.data) & 15) == 0;
-    // bucket <= 2048: whole bucket fits the register stash (4 groups/lane)
+    // bucket <= 2048: whole bucket fits the register stash (4 groups/lane);
+    // small power-of-two buckets (bucket < 512) pack several buckets per
+    // wave in their own kernel (kind 3, QuantSub)
     if (!fb && aligned && s.n >= s.bucket && s.bucket <= 2048) {
-      groups[{s.bits, 0}].push_back(Ent{&s, nullptr, flags});
+      const int ng = s.bucket >> 3;
+      const bool small_pow2 = ng < 64 && (ng & (ng - 1)) == 0;
+      groups[{s.bits, small_pow2 ? 3 : 0}].push_back(Ent{&s, nullptr, flags});
       if (!s.skip_incomplete && (s.n % s.bucket) != 0)
         groups[{s.bits, 1}].push_back(Ent{&s, nullptr,
                                           flags | kFlagTailOnly});
@@ -459,7 +463,7 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
       qd[i] = QuantDesc{s.data, out_base + s.comp_off, list[i].fb, s.n,
                         s.bucket, list[i].flags};
       int64_t nb;
-      if (kind == 0) {
+      if (kind == 0 || kind == 3) {
         nb = s.n / s.bucket;  // full buckets only
         any_residual |= s.skip_incomplete && (s.n % s.bucket) != 0;
       } else if (list[i].flags & kFlagTailOnly) {
@@ -475,7 +479,10 @@ void Engine::run_quantize(const std::vector<Slice>& slices, uint8_t* out_base,
     char* dev = (char*)ring_.commit(desc_bytes + cum_bytes, stream);
     auto* ddesc = reinterpret_cast<QuantDesc*>(dev);
     auto* dcum = reinterpret_cast<int64_t*>(dev + desc_bytes);
-    if (kind == 0) {
+    if (kind == 3) {
+      launch_quantize_sub(ddesc, dcum, nsl, cum[nsl], dt, bits, seed_++,
+                          stochastic, stream, any_residual);
+    } else if (kind == 0) {
       int max_gpl = 1;
       for (const auto& e : list)
         max_gpl = std::max(max_gpl,

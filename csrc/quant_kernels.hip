@@ -454,6 +454,121 @@ struct QuantRun {
 
 };
 
+// Sub-wave quantize for SMALL buckets (bucket/8 = L lanes per bucket, a
+// power of two < 64): the wave-per-bucket layout leaves (64-L) lanes idle
+// and serializes 64/L x more buckets per wave; here the wave processes
+// W = 64/L buckets at once, reducing each L-lane segment with the DPP
+// ladder truncated at L and broadcasting the segment result back with one
+// ds_bpermute per value.  Bucket 256 measured 1.3-2.4 TB/s on the
+// wave-per-bucket path; this recovers most of the bucket-1024 rate.
+template <int L>
+__device__ __forceinline__ void seg_minmax(float& vmin, float& vmax,
+                                           int lane) {
+  if constexpr (L > 1) {
+    int t;
+#define CGX_DPP_SEG(CTRL, OP, IDENT)                                       \
+  t = __builtin_amdgcn_update_dpp(IDENT, __builtin_bit_cast(int, v),       \
+                                  (CTRL), 0xf, 0xf, false);                \
+  v = OP(v, __builtin_bit_cast(float, t));
+    {
+      const int iid = __builtin_bit_cast(int, INFINITY);
+      float v = vmin;
+      CGX_DPP_SEG(0x111, fminf, iid)
+      if constexpr (L >= 4) CGX_DPP_SEG(0x112, fminf, iid)
+      if constexpr (L >= 8) CGX_DPP_SEG(0x114, fminf, iid)
+      if constexpr (L >= 16) CGX_DPP_SEG(0x118, fminf, iid)
+      if constexpr (L >= 32) CGX_DPP_SEG(0x142, fminf, iid)
+      vmin = v;
+    }
+    {
+      const int iid = __builtin_bit_cast(int, -INFINITY);
+      float v = vmax;
+      CGX_DPP_SEG(0x111, fmaxf, iid)
+      if constexpr (L >= 4) CGX_DPP_SEG(0x112, fmaxf, iid)
+      if constexpr (L >= 8) CGX_DPP_SEG(0x114, fmaxf, iid)
+      if constexpr (L >= 16) CGX_DPP_SEG(0x118, fmaxf, iid)
+      if constexpr (L >= 32) CGX_DPP_SEG(0x142, fmaxf, iid)
+      vmax = v;
+    }
+#undef CGX_DPP_SEG
+    // segment result sits at lane (s+1)*L-1; broadcast it to the segment
+    const int src = ((lane / L) * L + (L - 1)) * 4;
+    vmin = __builtin_bit_cast(
+        float, __builtin_amdgcn_ds_bpermute(src,
+                                            __builtin_bit_cast(int, vmin)));
+    vmax = __builtin_bit_cast(
+        float, __builtin_amdgcn_ds_bpermute(src,
+                                            __builtin_bit_cast(int, vmax)));
+  }
+}
+
+template <typename T, int BITS, int L>
+struct QuantSub {
+  using R = typename RawOf<T>::type;
+  const T* __restrict__ base;
+  R* __restrict__ meta;
+  uint8_t* __restrict__ packed;
+  int lane;
+  float divisor;
+  uint64_t seed;
+  int stochastic;
+  int slice_idx;
+
+  // buckets lb0, lb0+nw, ..., count of them; W = 64/L per wave iteration
+  __device__ void run(int64_t lb0, int64_t count, int64_t nw) const {
+    constexpr int W = kWave / L;
+    const int s = lane / L;   // my segment = which of the W buckets
+    const int li = lane % L;  // my 8-elem group within the bucket
+    for (int64_t i = 0; i < count; i += W) {
+      const bool active = (i + s) < count;
+      const int64_t lb = lb0 + (i + s) * nw;
+      uint32_t r[8];
+      float lmin = INFINITY, lmax = -INFINITY;
+      if (active) {
+        const T* in = base + lb * (int64_t)(L * 8) + li * 8;
+        load8<T>(in, true, r);
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          const float f = raw2f<T>(r[j]);
+          lmin = fminf(lmin, f);
+          lmax = fmaxf(lmax, f);
+        }
+      }
+      seg_minmax<L>(lmin, lmax, lane);
+      if (!active) continue;
+      const uint32_t unit_raw = f2raw<T>((lmax - lmin) / divisor);
+      const float unitf = raw2f<T>(unit_raw);
+      const float minf = lmin;
+      if (li == 0) {
+        meta[2 * lb] = static_cast<R>(unit_raw);
+        meta[2 * lb + 1] = static_cast<R>(f2raw<T>(lmin));
+      }
+      const int64_t g = lb * L + li;  // global group index in the slice
+      using acc_t =
+          typename std::conditional<(BITS <= 4), uint32_t, uint64_t>::type;
+      acc_t value = 0;
+      if (unitf >= kEps) {
+        const float rinv = 1.0f / unitf;
+        const uint64_t pr =
+            stochastic
+                ? rand_pack(seed, (static_cast<uint64_t>(slice_idx) << 44) |
+                                      static_cast<uint64_t>(g))
+                : 0;
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          const float rnd = stochastic ? rand_lane(pr, j) : 0.5f;
+          const float dd = (raw2f<T>(r[j]) - minf) * rinv + rnd;
+          const uint32_t level =
+              static_cast<uint32_t>(fminf(floorf(dd), divisor));
+          value |=
+              static_cast<acc_t>(level & ((1u << BITS) - 1)) << (j * BITS);
+        }
+      }
+      store_bytes(packed + g * BITS, static_cast<uint64_t>(value), BITS);
+    }
+  }
+};
+
 // Lean fast-path kernel (see compress.h launch_quantize_fast): only the
 // QuantRun register path is instantiated, so the register allocator is not
 // forced to the generic/EF paths' footprint (k_quantize<float,4,true>
@@ -510,6 +625,57 @@ __global__ __launch_bounds__(kThreads) void k_quantize_fast(
       }
     }
 #undef CGX_QRUNF
+    b += count * nw;
+  }
+}
+
+// Dedicated small-bucket kernel: ONLY QuantSub is instantiated (folding it
+// into k_quantize_fast measured VGPR 53 -> 107/143, regressing the large
+// buckets -- the same register-allocation coupling the lean split fixed).
+template <typename T, int BITS>
+__global__ __launch_bounds__(kThreads) void k_quantize_sub(
+    const QuantDesc* __restrict__ descs, const int64_t* __restrict__ cum,
+    int nslices, int64_t total_buckets, uint64_t seed, int stochastic) {
+  using R = typename RawOf<T>::type;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wid = static_cast<int64_t>(blockIdx.x) * (kThreads / kWave) +
+                      (threadIdx.x / kWave);
+  const int64_t nw = static_cast<int64_t>(gridDim.x) * (kThreads / kWave);
+  constexpr float divisor = static_cast<float>((1 << BITS) - 1);
+  int64_t b = wid;
+  while (b < total_buckets) {
+    int lo = 0, hi = nslices;
+    while (hi - lo > 1) {
+      const int mid = (lo + hi) >> 1;
+      if (cum[mid] <= b) lo = mid; else hi = mid;
+    }
+    const QuantDesc d = descs[lo];
+    const bool skip = (d.flags & kFlagSkipIncomplete) != 0;
+    const int64_t nb_slice =
+        skip ? d.n / d.bucket
+             : (d.n + d.bucket - 1) / static_cast<int64_t>(d.bucket);
+    const int64_t nb_full = cum[lo + 1] - cum[lo];
+    const int64_t lb = b - cum[lo];
+    const int ngroups = d.bucket >> 3;
+    const int64_t count = (nb_full - lb + nw - 1) / nw;
+#define CGX_QSUB(LV)                                                       \
+  do {                                                                     \
+    QuantSub<T, BITS, LV> q{reinterpret_cast<const T*>(d.in),              \
+                            reinterpret_cast<R*>(d.out),                   \
+                            reinterpret_cast<uint8_t*>(d.out) +            \
+                                2 * sizeof(R) * nb_slice,                  \
+                            lane, divisor, seed, stochastic, lo};          \
+    q.run(lb, count, nw);                                                  \
+  } while (0)
+    switch (ngroups) {
+      case 1: CGX_QSUB(1); break;
+      case 2: CGX_QSUB(2); break;
+      case 4: CGX_QSUB(4); break;
+      case 8: CGX_QSUB(8); break;
+      case 16: CGX_QSUB(16); break;
+      default: CGX_QSUB(32); break;
+    }
+#undef CGX_QSUB
     b += count * nw;
   }
 }
@@ -1547,6 +1713,26 @@ void launch_quantize_fast(const QuantDesc* descs, const int64_t* cum,
                            dim3(kThreads), 0, stream, descs, cum, nslices,
                            total_buckets, seed, (int)stochastic);
       }
+    }
+    if (any_residual) {
+      hipLaunchKernelGGL((k_residual_q<T>), dim3(32), dim3(kThreads), 0,
+                         stream, descs, nslices, bits);
+    }
+  }));
+}
+
+void launch_quantize_sub(const QuantDesc* descs, const int64_t* cum,
+                         int nslices, int64_t total_buckets, DType dt,
+                         int bits, uint64_t seed, bool stochastic,
+                         hipStream_t stream, bool any_residual) {
+  if (nslices <= 0) return;
+  const int grid = grid_for(std::max<int64_t>(total_buckets, 1),
+                            kThreads / kWave);
+  CGX_DISPATCH_T(dt, CGX_DISPATCH_BITS(bits, {
+    if (total_buckets > 0) {
+      hipLaunchKernelGGL((k_quantize_sub<T, BITS>), dim3(grid),
+                         dim3(kThreads), 0, stream, descs, cum, nslices,
+                         total_buckets, seed, (int)stochastic);
     }
     if (any_residual) {
       hipLaunchKernelGGL((k_residual_q<T>), dim3(32), dim3(kThreads), 0,
