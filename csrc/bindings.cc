@@ -98,6 +98,87 @@ void py_loopback_broadcast(std::vector<at::Tensor> tensors, int64_t root) {
   });
 }
 
+// Hierarchical flow on one GPU (backend.cc ProcessGroupCGX::allreduce
+// `hier` branch): ws = n_nodes * local_size ranks; each "node" gets an
+// intra hub, the node leaders share a cross hub.  Per rank: intra-node
+// compressed allreduce, cross-node reduction on the leader, intra
+// broadcast of the result — the registry resolved ONCE for all engines.
+void py_loopback_hierarchical(std::vector<at::Tensor> buckets,
+                              int64_t local_size) {
+  const int ws = (int)buckets.size();
+  TORCH_CHECK(local_size >= 1 && ws % local_size == 0 && ws >= 2,
+              "loopback_hierarchical: world must be n_nodes*local_size");
+  const int nodes = ws / (int)local_size;
+  TORCH_CHECK(nodes >= 2, "loopback_hierarchical: need >= 2 nodes");
+  for (const auto& t : buckets) {
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                    t.numel() == buckets[0].numel() &&
+                    t.device() == buckets[0].device(),
+                "loopback_hierarchical: matching CUDA tensors required");
+  }
+  const int dev = buckets[0].device().index();
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(dev);
+  Registry::BucketInfo info;
+  const bool matched =
+      Registry::get().next(buckets[0].numel(), nullptr, &info);
+  std::vector<std::unique_ptr<LoopbackHub>> intra_hubs;
+  for (int nd = 0; nd < nodes; nd++)
+    intra_hubs.push_back(std::make_unique<LoopbackHub>((int)local_size));
+  LoopbackHub cross_hub(nodes);
+  std::vector<std::unique_ptr<Engine>> intra_eng;    // per rank
+  std::vector<std::unique_ptr<LoopbackTransport>> intra_tr;
+  std::vector<std::unique_ptr<Engine>> cross_eng;    // per node (leader)
+  std::vector<std::unique_ptr<LoopbackTransport>> cross_tr;
+  std::vector<hipStream_t> qs(ws, nullptr);
+  for (int r = 0; r < ws; r++) {
+    const int nd = r / (int)local_size;
+    const int lr = r % (int)local_size;
+    intra_eng.push_back(std::make_unique<Engine>(lr, (int)local_size));
+    intra_tr.push_back(
+        std::make_unique<LoopbackTransport>(intra_hubs[nd].get(), lr));
+    CGX_HIP_CHECK(hipStreamCreateWithFlags(&qs[r], hipStreamNonBlocking));
+  }
+  for (int nd = 0; nd < nodes; nd++) {
+    cross_eng.push_back(
+        std::make_unique<Engine>(nd, nodes, /*is_cross=*/true));
+    cross_tr.push_back(std::make_unique<LoopbackTransport>(&cross_hub, nd));
+  }
+  std::vector<std::exception_ptr> errs(ws);
+  {
+    py::gil_scoped_release nogil;
+    std::vector<std::thread> th;
+    for (int r = 0; r < ws; r++) {
+      th.emplace_back([&, r] {
+        try {
+          c10::hip::HIPGuardMasqueradingAsCUDA g(dev);
+          const int nd = r / (int)local_size;
+          const int lr = r % (int)local_size;
+          hipStream_t fin = intra_eng[r]->allreduce(
+              buckets[r], intra_tr[r].get(), qs[r], &info, matched);
+          if (lr == 0) {
+            fin = cross_eng[nd]->allreduce(buckets[r], cross_tr[nd].get(),
+                                           fin, &info, matched);
+          }
+          fin = intra_eng[r]->broadcast(buckets[r], /*root=*/0,
+                                        intra_tr[r].get(), fin);
+          CGX_HIP_CHECK(hipStreamSynchronize(fin));
+          CGX_HIP_CHECK(hipStreamSynchronize(qs[r]));
+        } catch (...) {
+          errs[r] = std::current_exception();
+          for (auto& h : intra_hubs) h->abort();
+          cross_hub.abort();
+        }
+      });
+    }
+    for (auto& t : th) t.join();
+  }
+  (void)hipDeviceSynchronize();
+  for (auto s : qs)
+    if (s) (void)hipStreamDestroy(s);
+  for (int r = 0; r < ws; r++)
+    if (errs[r]) std::rethrow_exception(errs[r]);
+}
+
 DType dtype_arg(const at::Tensor& t) { return dtype_of(t); }
 
 // Compress a 1-D CUDA tensor; returns the uint8 compressed buffer
@@ -407,6 +488,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("root") = 0,
         "Run the production Engine::broadcast at world_size=len(tensors) on "
         "one GPU via the loopback transport.");
+  m.def("loopback_hierarchical", &cgx::py_loopback_hierarchical,
+        py::arg("buckets"), py::arg("local_size"),
+        "Run the production hierarchical flow (intra allreduce -> leader "
+        "cross reduction -> intra broadcast) for n_nodes*local_size ranks "
+        "on one GPU via loopback transports.");
   m.def("quantize", &cgx::py_quantize, py::arg("x"), py::arg("bits"),
         py::arg("bucket_size"), py::arg("stochastic") = false,
         py::arg("seed") = 0, py::arg("skip_incomplete") = false,
