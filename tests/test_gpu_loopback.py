@@ -331,3 +331,50 @@ def test_sra_low_precision_bitwise(_env, dtype):
     _C.loopback_allreduce(buckets)
     for r in range(ws):
         assert torch.equal(buckets[r].cpu(), sim[r]), (dtype, r)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_skip_incomplete_equal_inputs_exact(_env, ws):
+    """CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS at multi-rank: the trailing
+    partial bucket travels as raw values; equal inputs reduce exactly."""
+    from torch_cgx_amd import _C
+    _cfg(_env, 4, 512)
+    os.environ["CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS"] = "1"
+    try:
+        n = 10_000  # 19 full buckets + 272 raw residuals
+        buckets = [torch.full((n,), 1.5, device=_dev()) for _ in range(ws)]
+        _C.loopback_allreduce(buckets)
+        expected = torch.full((n,), 1.5 * ws, device=_dev())
+        for b in buckets:
+            assert torch.equal(b, expected)
+    finally:
+        os.environ.pop("CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS", None)
+
+
+def test_fuzz_registered_layers_vs_sim(_env):
+    """Randomized multi-layer registered buckets (odd sizes/offsets hit the
+    unaligned and tail paths) at random world sizes, bitwise vs the CPU
+    simulation."""
+    from torch_cgx_amd import _C
+    import numpy as np
+    rng = np.random.default_rng(17)
+    _cfg(_env, 32, 512)
+    for trial in range(6):
+        ws = int(rng.choice([2, 3, 4, 8]))
+        nl = int(rng.integers(1, 5))
+        layers = [int(rng.integers(30, 4000)) for _ in range(nl)]
+        cfgs = [(int(rng.choice([2, 4, 8])),
+                 int(rng.choice([64, 512, 1024]))) for _ in range(nl)]
+        _C.clear_registry()
+        for li, (numel, (b, bs)) in enumerate(zip(layers, cfgs)):
+            _C.register_layer(0, li, numel, b, bs)
+        n = sum(layers)
+        torch.manual_seed(trial)
+        cpu = [torch.randn(n) for _ in range(ws)]
+        sim = sra_sim.sra_allreduce([t.clone() for t in cpu], layers, cfgs)
+        buckets = [t.to(_dev()) for t in cpu]
+        _C.loopback_allreduce(buckets)
+        for r in range(ws):
+            assert torch.equal(buckets[r].cpu(), sim[r]), (
+                trial, ws, layers, cfgs, r)
+        _C.clear_registry()
