@@ -146,3 +146,22 @@ def _base_variants(rank, ws):
 
 def test_base_collectives_cpu():
     run_dist(_base_variants, 2)
+
+
+def _subgroup_worker(rank, ws):
+    # dist.new_group on the cgx backend (the reference exposed group names
+    # "for FSDP"; here subgroups get their own prefixed store + delegate)
+    import torch
+    g = dist.new_group([0, 1])
+    t = torch.full((10,), float(rank + 1))
+    if rank in (0, 1):
+        dist.all_reduce(t, group=g)
+        assert torch.equal(t, torch.full((10,), 3.0)), t
+    else:
+        assert torch.equal(t, torch.full((10,), float(rank + 1)))
+    dist.barrier()
+
+
+def test_subgroup_cpu_ws3():
+    from dist_utils import run_dist
+    run_dist(_subgroup_worker, 3)
