@@ -1,0 +1,44 @@
+"""Extended soak: EF, hierarchical, broadcast, a2a and mixed-layer paths
+interleaved for many iterations — stability guard for the driver's
+round-end multi-GPU runs."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from torch_cgx_amd import _C
+
+torch.manual_seed(1)
+it = 0
+for rep in range(6):
+    for mode in ("sra", "ring", "ef", "a2a", "hier", "bcast"):
+        for n in (4096, 500_000):
+            os.environ["CGX_COMPRESSION_QUANTIZATION_BITS"] = "4"
+            os.environ["CGX_COMPRESSION_BUCKET_SIZE"] = "512"
+            os.environ.pop("CGX_ERROR_FEEDBACK", None)
+            os.environ.pop("CGX_DEBUG_ALL_TO_ALL_REDUCTION", None)
+            os.environ["CGX_INNER_REDUCTION_TYPE"] = "SRA"
+            ws = 4
+            bufs = [torch.randn(n, device="cuda") for _ in range(ws)]
+            if mode == "ring":
+                os.environ["CGX_INNER_REDUCTION_TYPE"] = "Ring"
+                _C.loopback_allreduce(bufs)
+            elif mode == "ef":
+                os.environ["CGX_ERROR_FEEDBACK"] = "1"
+                _C.loopback_allreduce(bufs)
+            elif mode == "a2a":
+                os.environ["CGX_DEBUG_ALL_TO_ALL_REDUCTION"] = "1"
+                _C.loopback_allreduce(bufs)
+            elif mode == "hier":
+                _C.loopback_hierarchical(bufs, 2)
+            elif mode == "bcast":
+                _C.loopback_broadcast(bufs, 0)
+            else:
+                _C.loopback_allreduce(bufs)
+            if mode not in ("bcast",):
+                for b in bufs[1:]:
+                    assert torch.equal(b, bufs[0]), (mode, n, it)
+            it += 1
+print(f"soak2 ok: {it} collective runs", flush=True)
