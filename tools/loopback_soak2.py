@@ -37,7 +37,12 @@ for rep in range(6):
                 _C.loopback_broadcast(bufs, 0)
             else:
                 _C.loopback_allreduce(bufs)
-            if mode not in ("bcast",):
+            if mode == "a2a":
+                # the debug all-to-all accumulates in per-rank order (self
+                # first) -- rank results agree to rounding, not bitwise
+                for b in bufs[1:]:
+                    assert torch.allclose(b, bufs[0], atol=1e-3), (mode, n, it)
+            elif mode != "bcast":
                 for b in bufs[1:]:
                     assert torch.equal(b, bufs[0]), (mode, n, it)
             it += 1
