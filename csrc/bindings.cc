@@ -92,6 +92,31 @@ void py_loopback_allreduce(std::vector<at::Tensor> buckets) {
   });
 }
 
+// Multiple allreduce steps on PERSISTENT engines (one set of engines for
+// all steps, like a training run): steps[s][r] is rank r's bucket at step
+// s, reduced in place.  This is what lets error-feedback residuals carry
+// across steps on hardware.
+void py_loopback_allreduce_multi(
+    std::vector<std::vector<at::Tensor>> steps) {
+  TORCH_CHECK(!steps.empty(), "loopback_allreduce_multi: no steps");
+  const int ws = (int)steps[0].size();
+  const int64_t numel = steps[0][0].numel();
+  for (auto& st : steps) {
+    TORCH_CHECK((int)st.size() == ws && st[0].numel() == numel,
+                "loopback_allreduce_multi: ragged steps");
+  }
+  Registry::BucketInfo info;
+  const bool matched = Registry::get().next(numel, nullptr, &info);
+  run_loopback(steps[0],
+               [&](Engine& e, Transport* tr, hipStream_t qs, int r) {
+                 hipStream_t fin = qs;
+                 for (size_t s = 0; s < steps.size(); s++) {
+                   fin = e.allreduce(steps[s][r], tr, qs, &info, matched);
+                 }
+                 return fin;
+               });
+}
+
 void py_loopback_broadcast(std::vector<at::Tensor> tensors, int64_t root) {
   run_loopback(tensors, [&](Engine& e, Transport* tr, hipStream_t qs, int r) {
     return e.broadcast(tensors[r], (int)root, tr, qs);
@@ -488,6 +513,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("root") = 0,
         "Run the production Engine::broadcast at world_size=len(tensors) on "
         "one GPU via the loopback transport.");
+  m.def("loopback_allreduce_multi", &cgx::py_loopback_allreduce_multi,
+        py::arg("steps"),
+        "Multiple allreduce steps on persistent engines (steps[s][r] = rank "
+        "r's bucket at step s, reduced in place); error-feedback residuals "
+        "carry across steps.");
   m.def("loopback_hierarchical", &cgx::py_loopback_hierarchical,
         py::arg("buckets"), py::arg("local_size"),
         "Run the production hierarchical flow (intra allreduce -> leader "
