@@ -695,6 +695,20 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::alltoall(
                     });
 }
 
+namespace {
+// CGX_P2P_ANYSOURCE=1 enables MPI_ANY_SOURCE-style GPU receives (reference
+// ProcessGroupCGX.cc:765-785): RCCL p2p needs a known peer, so every GPU
+// send also posts a tiny gloo "announcement" carrying the source rank, and
+// every GPU receive consumes one announcement first (recvAnysource learns
+// its peer from it).  Off by default — it adds a host-side gloo message to
+// every GPU send, and mixing announced and plain p2p would desynchronize
+// the pairing, so the flag must be set consistently on ALL ranks.
+bool p2p_anysource() {
+  static const bool v = env_flag("CGX_P2P_ANYSOURCE");
+  return v;
+}
+}  // namespace
+
 c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::send(
     std::vector<at::Tensor>& tensors, int dstRank, int tag) {
   check_single(tensors);
@@ -702,6 +716,12 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::send(
   if (!t.is_cuda()) {
     TORCH_CHECK(cpu_, "cgx: no CPU delegate backend available");
     return cpu_->send(tensors, dstRank, tag);
+  }
+  if (p2p_anysource()) {
+    TORCH_CHECK(cpu_, "cgx: CGX_P2P_ANYSOURCE requires the CPU delegate");
+    std::vector<at::Tensor> ann{
+        at::full({1}, rank_, at::TensorOptions().dtype(at::kInt))};
+    cpu_->send(ann, dstRank, tag)->wait();
   }
   return collective(tensors, t.device(), c10d::OpType::SEND,
                     [this, t, dstRank](hipStream_t s) {
@@ -719,6 +739,12 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::recv(
     TORCH_CHECK(cpu_, "cgx: no CPU delegate backend available");
     return cpu_->recv(tensors, srcRank, tag);
   }
+  if (p2p_anysource()) {
+    // consume this sender's announcement to keep the pairing in lockstep
+    std::vector<at::Tensor> ann{
+        at::zeros({1}, at::TensorOptions().dtype(at::kInt))};
+    cpu_->recv(ann, srcRank, tag)->wait();
+  }
   return collective(tensors, t.device(), c10d::OpType::RECV,
                     [this, t, srcRank](hipStream_t s) {
                       CGX_NCCL_CHECK(ncclRecv(t.data_ptr(), t.numel(),
@@ -734,9 +760,28 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::recvAnysource(
     TORCH_CHECK(cpu_, "cgx: no CPU delegate backend available");
     return cpu_->recvAnysource(tensors, tag);
   }
-  TORCH_CHECK(false,
-              "cgx: recvAnysource is not supported for GPU tensors (RCCL "
-              "p2p requires a known source rank)");
+  TORCH_CHECK(p2p_anysource(),
+              "cgx: recvAnysource for GPU tensors requires "
+              "CGX_P2P_ANYSOURCE=1 on every rank (RCCL p2p needs a known "
+              "source; the flag adds a gloo source announcement to each "
+              "GPU send)");
+  at::Tensor t = tensors[0];
+  // learn the source from the announcement, then RCCL-receive from it
+  std::vector<at::Tensor> ann{
+      at::zeros({1}, at::TensorOptions().dtype(at::kInt))};
+  auto w = cpu_->recvAnysource(ann, tag);
+  w->wait();
+  const int src = w->sourceRank() >= 0
+                      ? w->sourceRank()
+                      : ann[0].item<int>();
+  TORCH_CHECK(src >= 0 && src < size_ && src != rank_,
+              "cgx: bad announced source rank ", src);
+  return collective(tensors, t.device(), c10d::OpType::RECVANYSOURCE,
+                    [this, t, src](hipStream_t s) {
+                      CGX_NCCL_CHECK(ncclRecv(t.data_ptr(), t.numel(),
+                                              nccl_dtype(t), src, comm_,
+                                              s));
+                    });
 }
 
 c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::barrier(

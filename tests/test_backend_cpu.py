@@ -165,3 +165,20 @@ def _subgroup_worker(rank, ws):
 def test_subgroup_cpu_ws3():
     from dist_utils import run_dist
     run_dist(_subgroup_worker, 3)
+
+
+def _anysource_worker(rank, ws):
+    import torch
+    t = torch.zeros(4)
+    if rank == 1:
+        dist.send(torch.full((4,), 7.0), dst=0)
+    elif rank == 0:
+        src = dist.recv(t, src=None)  # recvAnysource via the gloo delegate
+        assert src == 1, src
+        assert torch.equal(t, torch.full((4,), 7.0))
+    dist.barrier()
+
+
+def test_recv_anysource_cpu():
+    from dist_utils import run_dist
+    run_dist(_anysource_worker, 2)
