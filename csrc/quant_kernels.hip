@@ -4,14 +4,29 @@
 //  * wavefront = 64 lanes; one WAVE per quantization bucket — for buckets up
 //    to 2048 every lane's packs stay in registers between the meta and
 //    encode phases (single HBM read), the min/max reduction is a DPP row-op
-//    sequence (register path; no LDS, no ds_bpermute), and 16-bit dtypes
-//    reduce with packed v_pk_min/max.
+//    sequence; 16-bit dtypes reduce with packed v_pk_min/max; small
+//    power-of-two buckets (< 512) pack 64/(bucket/8) buckets per wave with
+//    segmented DPP reductions (QuantSub).
+//  * REGISTER-ISOLATED kernel family (the round-2 lesson, measured three
+//    times over): the register allocator sizes a kernel for its worst path,
+//    so the hot paths live in their own lean launches —
+//      k_quantize_fast  (53-89 VGPR; QuantRun, compile-time groups/lane)
+//      k_quantize_sub   (small buckets, multi-bucket-per-wave)
+//      k_quantize       (generic: error feedback, unaligned, partial-bucket
+//                        tails re-entering via kFlagTailOnly, bucket%8!=0
+//                        meta pass)
+//      k_dequant_fast   (56/33 VGPR; branch-free, incremental bucket index)
+//      k_dequant        (generic + ragged tails via kFlagTailOnly)
+//    The host (engine.cc run_quantize/run_dequant, bindings) routes slices.
 //  * memory-bound workload: 16-byte vectorized loads/stores wherever the
 //    slice base is 16B-aligned; decode uses 4 elems/thread for fp32 so every
-//    store is one coalesced float4; grid-stride loops sized >> 256 CUs.
+//    store is one coalesced float4; no integer division or per-access
+//    alignment branches in any hot loop (incremental bucket trackers,
+//    compile-time alignment/accumulate template flags).
 //  * stochastic rounding via a stateless splitmix64 hash per pack (no RNG
-//    state buffers, deterministic given the per-launch seed); optional
-//    error-feedback residual update fused into the encode.
+//    state buffers, deterministic given the per-launch seed; the hash key is
+//    the GLOBAL pack index, so the kernel split is byte-transparent);
+//    optional error-feedback residual update fused into the encode.
 //
 // Wire-format parity with the reference implementation is defined by
 // torch_cgx_amd/ops/golden.py (see reference
