@@ -457,7 +457,8 @@ struct QuantRun {
   }
 
   // count buckets at lb0, lb0+nw, ...  Single stash bank (used by the
-  // MAXGT=4 kernel variant, where a second bank would cost occupancy).
+  // MAXGT>=2 kernel variants, where a second bank costs occupancy — the
+  // two-bank pipeline was measured and rejected twice, see RESULTS.md).
   __device__ __forceinline__ void run(int64_t lb0, int64_t count,
                                       int64_t nw) const {
     Stash A;
