@@ -16,6 +16,11 @@ import os
 import sys
 import time
 
+# dmabuf IPC is the only mode the host driver supports; RCCL multi-process
+# tensor sharing fails with hipIpcGetMemHandle errors without this (must be
+# set before HIP initializes)
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 import torch

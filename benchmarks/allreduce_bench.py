@@ -19,6 +19,8 @@ With WORLD_SIZE==1 it times the quantize/dequantize kernels only.
 import argparse
 import json
 import os
+
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 import sys
 import time
 

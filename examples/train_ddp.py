@@ -10,6 +10,8 @@ Launch:  torchrun --standalone --nproc-per-node 8 examples/train_ddp.py \
 
 import argparse
 import os
+
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 import sys
 import time
 

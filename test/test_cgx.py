@@ -11,6 +11,8 @@ between sweeps — launched with torchrun instead of mpirun:
 """
 
 import os
+
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 import unittest
 
 import numpy as np
