@@ -721,7 +721,16 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::send(
     TORCH_CHECK(cpu_, "cgx: CGX_P2P_ANYSOURCE requires the CPU delegate");
     std::vector<at::Tensor> ann{
         at::full({1}, rank_, at::TensorOptions().dtype(at::kInt))};
-    cpu_->send(ann, dstRank, tag)->wait();
+    // fire-and-forget: waiting here would host-block until the peer posts
+    // its receive (gloo p2p is unbuffered), deadlocking send/send-then-
+    // recv/recv patterns that plain async sends allow.  The Work (and the
+    // announcement tensor it references) stays queued until completed.
+    {
+      std::lock_guard<std::mutex> g(ann_mu_);
+      pending_ann_.push_back(cpu_->send(ann, dstRank, tag));
+      while (!pending_ann_.empty() && pending_ann_.front()->isCompleted())
+        pending_ann_.pop_front();
+    }
   }
   return collective(tensors, t.device(), c10d::OpType::SEND,
                     [this, t, dstRank](hipStream_t s) {

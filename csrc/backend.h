@@ -22,6 +22,7 @@
 #include <torch/csrc/distributed/c10d/Work.hpp>
 
 #include <memory>
+#include <deque>
 #include <mutex>
 #include <optional>
 #include <vector>
@@ -141,6 +142,9 @@ class ProcessGroupCGX : public c10d::Backend {
   c10::intrusive_ptr<c10d::Backend> cpu_;
   ncclComm_t comm_ = nullptr;
   std::unique_ptr<RcclTransport> tr_;
+  // CGX_P2P_ANYSOURCE: in-flight gloo source announcements (fire-and-forget)
+  std::mutex ann_mu_;
+  std::deque<c10::intrusive_ptr<c10d::Work>> pending_ann_;
   int device_index_ = -1;
   std::unique_ptr<Engine> engine_;
   // hierarchical (multi-node) mode: intra-node SRA + cross-node reduction on
