@@ -182,3 +182,4 @@ def _anysource_worker(rank, ws):
 def test_recv_anysource_cpu():
     from dist_utils import run_dist
     run_dist(_anysource_worker, 2)
+

@@ -234,3 +234,28 @@ def test_error_feedback_repeated_use():
     _C.dequantize(one, out, bits, bucket, False)
     one_err = (out - x).abs().max().item()
     assert ef_err < one_err * 0.35, (ef_err, one_err)
+
+
+def test_multisource_dequant_throughput_floor():
+    """Regression guard for the SRA round-1 decode: 7-source accumulate of
+    64M fp32 4-bit must stream at >= 2.5 TB/s effective (round-2 measured
+    ~4 TB/s)."""
+    from torch_cgx_amd import _C
+    n, bits, bucket, nsrc = 64 << 20, 4, 1024, 7
+    x = torch.randn(n, device=_dev())
+    comp = _C.quantize(x, bits, bucket, True, 0)
+    multi = comp.unsqueeze(0).repeat(nsrc, 1).contiguous()
+    out = torch.empty_like(x)
+    for _ in range(3):
+        _C.dequantize_multi(multi, out, bits, bucket, True)
+    torch.cuda.synchronize()
+    import time
+    t0 = time.perf_counter()
+    reps = 10
+    for _ in range(reps):
+        _C.dequantize_multi(multi, out, bits, bucket, True)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / reps
+    eff = (n * 4 * 2 + nsrc * comp.numel()) / dt / 1e9
+    print(f"7-src dequant 64M fp32 4-bit: {dt*1e3:.2f} ms, {eff:.0f} GB/s")
+    assert eff > 2500, f"multi-source dequant too slow: {eff:.0f} GB/s"

@@ -412,3 +412,16 @@ def test_error_feedback_accumulates_across_steps(_env, ws):
     err_ef = run(True)
     err_no = run(False)
     assert err_ef < err_no * 0.5, (ws, err_ef, err_no)
+
+
+def test_plain_broadcast_bits32(_env):
+    """bits=32 -> Engine::broadcast takes the uncompressed transport path."""
+    from torch_cgx_amd import _C
+    _cfg(_env, 32, 512)
+    n = 5000
+    src = torch.randn(n)
+    tensors = [src.to(_dev()) if r == 0 else torch.zeros(n, device=_dev())
+               for r in range(4)]
+    _C.loopback_broadcast(tensors, 0)
+    for t in tensors:
+        assert torch.equal(t.cpu(), src)
