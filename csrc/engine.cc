@@ -858,10 +858,13 @@ void Engine::a2a_chunk(const std::vector<LayerView>& views, DType dt,
                        const EngineConfig& cfg) {
   // Debug brute-force reduction (reference AllReduceAlltoAllCompressed,
   // scatter_reduce_allgather.cc:269-306): every rank quantizes its ENTIRE
-  // chunk once and sends it to every peer; all ranks then decode the same
-  // ws compressed streams (self first, overwrite; peers accumulated) so the
-  // result is bit-identical across ranks.  ws x the wire bytes of SRA, but
-  // no partitioning/offset machinery in the fault surface.
+  // chunk once and sends it to every peer; every rank then decodes the same
+  // ws compressed streams (self first, then peers in rank order).  Each
+  // rank's data is quantized exactly once, but the per-rank ACCUMULATION
+  // order differs (self-first), so at ws>2 results agree to fp rounding,
+  // not bitwise — same as the reference's arrival-order accumulate.  ws x
+  // the wire bytes of SRA, but no partitioning machinery in the fault
+  // surface.
   TORCH_CHECK(!cfg.error_feedback,
               "cgx: CGX_DEBUG_ALL_TO_ALL_REDUCTION does not support "
               "CGX_ERROR_FEEDBACK");

@@ -224,7 +224,9 @@ class Engine {
   // reference scatter_reduce_allgather.cc:269-306): every rank quantizes its
   // ENTIRE chunk once, sends it to all peers, and every rank decodes the
   // same ws compressed streams — no partitioning, ws× the wire traffic, but
-  // removes the partition/offset machinery from the fault surface.
+  // removes the partition/offset machinery from the fault surface.  Ranks
+  // accumulate in different orders (self first), so at ws>2 results agree
+  // to fp rounding, not bitwise.
   void a2a_chunk(const std::vector<LayerView>& views, DType dt,
                  Transport* tr, hipStream_t qs, const EngineConfig& cfg);
   uint8_t* staging(int64_t bytes, hipStream_t user);     // single-stream path
