@@ -800,9 +800,11 @@ c10::intrusive_ptr<c10d::Work> ProcessGroupCGX::barrier(
   if (comm_) {
     std::lock_guard<std::mutex> lock(mu_);
     CGX_HIP_CHECK(hipStreamSynchronize(stream_->stream()));
-    if (engine_) {
-      CGX_HIP_CHECK(hipStreamSynchronize(engine_->comm_stream()));
-      CGX_HIP_CHECK(hipStreamSynchronize(engine_->deq_stream()));
+    for (Engine* e : {engine_.get(), intra_engine_.get(),
+                      cross_engine_.get()}) {
+      if (!e) continue;
+      CGX_HIP_CHECK(hipStreamSynchronize(e->comm_stream()));
+      CGX_HIP_CHECK(hipStreamSynchronize(e->deq_stream()));
     }
   }
   if (cpu_) {
