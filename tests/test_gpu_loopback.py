@@ -425,3 +425,46 @@ def test_plain_broadcast_bits32(_env):
     _C.loopback_broadcast(tensors, 0)
     for t in tensors:
         assert torch.equal(t.cpu(), src)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_dummy_compression_exact(_env, ws):
+    """CGX_DEBUG_DUMMY_COMPRESSION forces the uncompressed transport path:
+    the result must be the exact float sum."""
+    from torch_cgx_amd import _C
+    _cfg(_env, 4, 512)
+    os.environ["CGX_DEBUG_DUMMY_COMPRESSION"] = "1"
+    try:
+        torch.manual_seed(31)
+        n = 20_000
+        cpu = [torch.randn(n) for _ in range(ws)]
+        expected = sum(cpu)
+        buckets = [t.to(_dev()) for t in cpu]
+        _C.loopback_allreduce(buckets)
+        for b in buckets:
+            assert torch.allclose(b.cpu(), expected, atol=1e-4)
+    finally:
+        os.environ.pop("CGX_DEBUG_DUMMY_COMPRESSION", None)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_fake_ratio_reduces_prefix_only(_env, ws):
+    """CGX_COMPRESSION_FAKE_RATIO=0.5: only the first half of each chunk is
+    reduced (bandwidth experiment, intentionally lossy); the tail must be
+    left exactly as each rank's input."""
+    from torch_cgx_amd import _C
+    _cfg(_env, 4, 512)
+    os.environ["CGX_COMPRESSION_FAKE_RATIO"] = "0.5"
+    try:
+        n = 10_000
+        # equal inputs -> the reduced prefix is exactly ws*x; the unreduced
+        # tail stays x
+        buckets = [torch.full((n,), 2.0, device=_dev()) for _ in range(ws)]
+        _C.loopback_allreduce(buckets)
+        got = buckets[0].cpu()
+        n_reduced = int((got == 2.0 * ws).sum().item())
+        n_untouched = int((got == 2.0).sum().item())
+        assert n_reduced + n_untouched == n, (n_reduced, n_untouched)
+        assert abs(n_reduced - n // 2) < 64, n_reduced  # ~half, aligned
+    finally:
+        os.environ.pop("CGX_COMPRESSION_FAKE_RATIO", None)
