@@ -468,3 +468,25 @@ def test_fake_ratio_reduces_prefix_only(_env, ws):
         assert abs(n_reduced - n // 2) < 64, n_reduced  # ~half, aligned
     finally:
         os.environ.pop("CGX_COMPRESSION_FAKE_RATIO", None)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_skip_incomplete_matches_sim_bitwise(_env, ws):
+    """skip_incomplete at multi-rank, bitwise vs the extended CPU sim
+    (raw residual tails + quantized full buckets through both rounds)."""
+    from torch_cgx_amd import _C
+    bits, bucket = 4, 512
+    _cfg(_env, bits, bucket)
+    os.environ["CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS"] = "1"
+    try:
+        torch.manual_seed(41)
+        n = 3000
+        cpu = [torch.randn(n) for _ in range(ws)]
+        sim = sra_sim.sra_allreduce([t.clone() for t in cpu], [n],
+                                    [(bits, bucket)], skip=True)
+        buckets = [t.to(_dev()) for t in cpu]
+        _C.loopback_allreduce(buckets)
+        for r in range(ws):
+            assert torch.equal(buckets[r].cpu(), sim[r]), (ws, r)
+    finally:
+        os.environ.pop("CGX_COMPRESSION_SKIP_INCOMPLETE_BUCKETS", None)

@@ -32,12 +32,14 @@ from . import partition as P
 
 def compress_chunk(buf: torch.Tensor, layer_numels: Sequence[int],
                    layer_configs: Sequence[Tuple[int, int]],
-                   start: int, size: int, rand=0.5) -> torch.Tensor:
+                   start: int, size: int, rand=0.5,
+                   skip: bool = False) -> torch.Tensor:
     """Compress [start, start+size) of buf slice-by-slice -> uint8 tensor."""
     parts = []
     for (li, off, n) in P.layer_slices(layer_numels, start, size):
         bits, bucket = layer_configs[li]
-        parts.append(golden.quantize(buf[off: off + n], bits, bucket, rand))
+        parts.append(golden.quantize(buf[off: off + n], bits, bucket, rand,
+                                     skip_incomplete=skip))
     if not parts:
         return torch.zeros(0, dtype=torch.uint8)
     return torch.cat(parts)
@@ -46,13 +48,16 @@ def compress_chunk(buf: torch.Tensor, layer_numels: Sequence[int],
 def decompress_chunk(comp: torch.Tensor, buf: torch.Tensor,
                      layer_numels: Sequence[int],
                      layer_configs: Sequence[Tuple[int, int]],
-                     start: int, size: int, add: bool) -> None:
+                     start: int, size: int, add: bool,
+                     skip: bool = False) -> None:
     """Decompress a compress_chunk() stream into buf (overwrite or +=)."""
     pos = 0
     for (li, off, n) in P.layer_slices(layer_numels, start, size):
         bits, bucket = layer_configs[li]
-        nbytes = golden.buffer_size(n, buf.dtype, bits, bucket)
-        vals = golden.dequantize(comp[pos: pos + nbytes], n, buf.dtype, bits, bucket)
+        nbytes = golden.buffer_size(n, buf.dtype, bits, bucket,
+                                    skip_incomplete=skip)
+        vals = golden.dequantize(comp[pos: pos + nbytes], n, buf.dtype, bits,
+                                 bucket, skip_incomplete=skip)
         if add:
             buf[off: off + n] += vals
         else:
@@ -63,7 +68,7 @@ def decompress_chunk(comp: torch.Tensor, buf: torch.Tensor,
 def sra_allreduce(tensors: List[torch.Tensor],
                   layer_numels: Sequence[int],
                   layer_configs: Sequence[Tuple[int, int]],
-                  rand=0.5) -> List[torch.Tensor]:
+                  rand=0.5, skip: bool = False) -> List[torch.Tensor]:
     """Simulate the compressed SRA over len(tensors) ranks; returns results."""
     ws = len(tensors)
     dtype = tensors[0].dtype
@@ -78,17 +83,17 @@ def sra_allreduce(tensors: List[torch.Tensor],
             if p == k:
                 continue
             comp = compress_chunk(out[p], layer_numels, layer_configs,
-                                  offsets[k], sizes[k], rand)
+                                  offsets[k], sizes[k], rand, skip)
             decompress_chunk(comp, out[k], layer_numels, layer_configs,
-                             offsets[k], sizes[k], add=True)
+                             offsets[k], sizes[k], add=True, skip=skip)
 
     # self-quantize + round 2: allgather
     for k in range(ws):
         comp = compress_chunk(out[k], layer_numels, layer_configs,
-                              offsets[k], sizes[k], rand)
+                              offsets[k], sizes[k], rand, skip)
         for p in range(ws):
             decompress_chunk(comp, out[p], layer_numels, layer_configs,
-                             offsets[k], sizes[k], add=False)
+                             offsets[k], sizes[k], add=False, skip=skip)
 
     return [o.view(tensors[i].shape) for i, o in enumerate(out)]
 
