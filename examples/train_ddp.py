@@ -34,6 +34,10 @@ def main():
     p.add_argument("--quantization-bits", type=int, default=4)
     p.add_argument("--quantization-bucket-size", type=int, default=1024)
     p.add_argument("--dist-backend", default="cgx")
+    p.add_argument("--error-feedback", action="store_true",
+                   help="fold the quantization residual into the next step")
+    p.add_argument("--adaptive", action="store_true",
+                   help="step-scheduled bits: fp32 warmup -> 8 -> target")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -45,11 +49,20 @@ def main():
     model = resnet18(num_classes=100).cuda()
     model = nn.parallel.DistributedDataParallel(model,
                                                 device_ids=[local_rank])
-    state = torch_cgx_amd.CGXState(
-        None, layer_min_size=1024,
-        compression_params={"bits": args.quantization_bits,
-                            "bucket_size": args.quantization_bucket_size})
-    model.register_comm_hook(state, torch_cgx_amd.cgx_hook)
+    if args.adaptive:
+        state = torch_cgx_amd.AdaptiveCGXState(
+            None, schedule=[(0, 32), (10, 8), (30, args.quantization_bits)],
+            bucket_size=args.quantization_bucket_size,
+            error_feedback=args.error_feedback)
+        model.register_comm_hook(state, torch_cgx_amd.adaptive_cgx_hook)
+    else:
+        params = {"bits": args.quantization_bits,
+                  "bucket_size": args.quantization_bucket_size}
+        if args.error_feedback:
+            params["error_feedback"] = True
+        state = torch_cgx_amd.CGXState(None, layer_min_size=1024,
+                                       compression_params=params)
+        model.register_comm_hook(state, torch_cgx_amd.cgx_hook)
 
     opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
                           weight_decay=5e-4)
