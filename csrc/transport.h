@@ -56,8 +56,10 @@ struct Transport {
 
 class RcclTransport final : public Transport {
  public:
-  RcclTransport(ncclComm_t comm, int rank, int size)
-      : comm_(comm), rank_(rank), size_(size) {}
+  RcclTransport(ncclComm_t comm, int rank, int size) : comm_(comm) {
+    (void)rank;
+    (void)size;
+  }
   void exchange(const std::vector<Op>& sends, const std::vector<Op>& recvs,
                 hipStream_t stream) override;
   void broadcast(void* ptr, int64_t count, int esize, ncclDataType_t dt,
@@ -70,7 +72,6 @@ class RcclTransport final : public Transport {
 
  private:
   ncclComm_t comm_;
-  int rank_, size_;
 };
 
 // Shared rendezvous for N loopback transports in one process.  Each
