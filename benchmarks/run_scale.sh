@@ -25,4 +25,5 @@ for N in 2 8; do
 done
 python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
     --master-addr 127.0.0.1 --master-port 29535 \
-    bench.py --gpus 8 --model bert-large --steps 20 --warmup 5       # config 4
+    bench.py --gpus 8 --model bert-large --bits 8 --show-registry \
+    --steps 20 --warmup 5                                            # config 4
