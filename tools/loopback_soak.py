@@ -13,7 +13,7 @@ from torch_cgx_amd import _C
 
 torch.manual_seed(0)
 it = 0
-for rep in range(4):
+for rep in range(int(os.environ.get("SOAK_REPS", "4"))):
     for ws in (2, 3, 4, 8):
         for bits in (1, 4, 8):
             for n in (1000, 65536, 1_000_003):

@@ -12,7 +12,7 @@ from torch_cgx_amd import _C
 
 torch.manual_seed(1)
 it = 0
-for rep in range(6):
+for rep in range(int(os.environ.get("SOAK_REPS", "6"))):
     for mode in ("sra", "ring", "ef", "a2a", "hier", "bcast"):
         for n in (4096, 500_000):
             os.environ["CGX_COMPRESSION_QUANTIZATION_BITS"] = "4"
