@@ -330,7 +330,7 @@ void py_dequantize(at::Tensor comp, at::Tensor out, int64_t bits,
   hb.cum[0] = 0;
   hb.cum[1] = (nq + 7) / 8;
   // mirror the engine's launch split (run_dequant)
-  if (bucket_size % 8 == 0 && nq < (int64_t(1) << 28)) {
+  if (bucket_size % 8 == 0 && nq < (int64_t(1) << 31)) {
     auto dev = upload(hb);
     const char* devp = static_cast<const char*>(dev.data_ptr());
     launch_dequantize_fast(reinterpret_cast<const DequantDesc*>(devp), 1,
@@ -384,7 +384,7 @@ void py_dequantize_multi(at::Tensor comp, at::Tensor out, int64_t bits,
   auto stream =
       c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(out.device().index());
   const char* devp = static_cast<const char*>(dev.data_ptr());
-  if (bucket_size % 8 == 0 && n < (int64_t(1) << 28)) {
+  if (bucket_size % 8 == 0 && n < (int64_t(1) << 31)) {
     launch_dequantize_fast(reinterpret_cast<const DequantDesc*>(devp), 1,
                            hb.cum[1], dt, (int)bits, stream.stream());
     const bool ragged = elem_size(dt) == 4 ? (n & 3) != 0 : (n & 7) != 0;

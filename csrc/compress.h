@@ -112,7 +112,7 @@ void launch_dequantize_batch(const DequantDesc* descs, const int64_t* cum,
                              bool any_residual = false);
 
 // Lean dequant kernel: every slice has bucket % 8 == 0 and fits u32
-// indexing (n < 2^28); ragged tails (n % 4 fp32 / n % 8 16-bit) re-enter
+// indexing (n < 2^31; per-thread indices are 4/8-element units); ragged tails (n % 4 fp32 / n % 8 16-bit) re-enter
 // launch_dequantize_batch with kFlagTailOnly.
 void launch_dequantize_fast(const DequantDesc* descs, int nslices,
                             int64_t total_groups, DType dt, int bits,

@@ -513,7 +513,7 @@ void Engine::run_dequant(const std::vector<Slice>& slices,
     const int32_t flags = s.skip_incomplete ? kFlagSkipIncomplete : 0;
     const int64_t nq =
         s.skip_incomplete ? s.n / s.bucket * (int64_t)s.bucket : s.n;
-    const bool fast = (s.bucket % 8) == 0 && nq < (int64_t(1) << 28);
+    const bool fast = (s.bucket % 8) == 0 && nq < (int64_t(1) << 31);
     if (fast) {
       groups[{s.bits, 0}].push_back(Ent{&s, flags});
       const bool ragged = es == 4 ? (nq & 3) != 0 : (nq & 7) != 0;

@@ -1328,7 +1328,7 @@ __global__ __launch_bounds__(kThreads) void k_dequant(
     const bool oneb = (d.bucket & 7) == 0;
     const uint32_t B8 = oneb ? static_cast<uint32_t>(d.bucket >> 3) : 1u;
     const bool al16 = (reinterpret_cast<uintptr_t>(d.out) & 15) == 0;
-    const bool small = nq < (int64_t(1) << 28);
+    const bool small = nq < (int64_t(1) << 31);
     T* const out_base = reinterpret_cast<T*>(d.out);
     const uint8_t* const in0 = d.in + meta_bytes;  // packed base, source 0
 

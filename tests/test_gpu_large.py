@@ -46,7 +46,8 @@ def _check_windows(x, comp, n, dtype, bits, bucket):
 
 
 @pytest.mark.parametrize("n,dtype,bits,bucket", [
-    (300_000_000, torch.float32, 4, 1024),   # > 2^28: dequant small=false
+    (300_000_000, torch.float32, 4, 1024),   # 2^28 < n < 2^31: u32 fast path
+    (268_435_456, torch.float32, 4, 1024),   # exactly 2^28 (old u32 boundary)
     (2_200_000_000, torch.float16, 4, 1024),  # > 2^31 elements
 ])
 def test_huge_tensor_quantize_windows(n, dtype, bits, bucket):
