@@ -103,6 +103,50 @@ def _gpu_sra(tensors, layer_numels, layer_configs):
     return out
 
 
+def test_ws1_more_collectives_cuda():
+    """Every remaining collective entry point at ws=1 on the real RCCL
+    backend (Work/event machinery + the size-1 edge of each RCCL call)."""
+    dist = _init_ws1()
+    dev = _dev()
+    t = torch.arange(64.0, device=dev)
+
+    outs = [torch.zeros(64, device=dev)]
+    dist.all_gather(outs, t)
+    assert torch.equal(outs[0], t)
+
+    big = torch.zeros(64, device=dev)
+    dist.all_gather_into_tensor(big, t)
+    assert torch.equal(big, t)
+
+    g = [torch.zeros(64, device=dev)]
+    dist.gather(t, g, dst=0)
+    assert torch.equal(g[0], t)
+
+    sc = torch.zeros(64, device=dev)
+    dist.scatter(sc, [t.clone()], src=0)
+    assert torch.equal(sc, t)
+
+    r = t.clone()
+    dist.reduce(r, dst=0)
+    assert torch.equal(r, t)
+
+    rs = torch.zeros(64, device=dev)
+    dist.reduce_scatter(rs, [t.clone()])
+    assert torch.equal(rs, t)
+
+    rsb = torch.zeros(64, device=dev)
+    dist.reduce_scatter_tensor(rsb, t.clone())
+    assert torch.equal(rsb, t)
+
+    a2a = torch.zeros(64, device=dev)
+    dist.all_to_all_single(a2a, t)
+    assert torch.equal(a2a, t)
+
+    ac = [t.clone(), (2 * t).clone()]
+    dist.all_reduce_coalesced(ac)
+    assert torch.equal(ac[0], t) and torch.equal(ac[1], 2 * t)
+
+
 @pytest.mark.parametrize("ws", [2, 4, 8])
 @pytest.mark.parametrize("dtype", [torch.float32, torch.float16])
 def test_sra_one_device_matches_cpu_sim(ws, dtype):
