@@ -275,7 +275,7 @@ def test_compressed_broadcast(_env, ws):
         assert torch.equal(tensors[r].cpu(), expected), r
 
 
-@pytest.mark.parametrize("nodes,local", [(2, 2), (2, 4), (4, 2)])
+@pytest.mark.parametrize("nodes,local", [(2, 2), (2, 4), (4, 2), (2, 8)])
 def test_hierarchical_matches_composed_sim(_env, nodes, local):
     """The full hierarchical flow (intra SRA -> leader cross reduction ->
     compressed intra broadcast) on hardware vs a composed CPU simulation.
