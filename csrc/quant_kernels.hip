@@ -1084,7 +1084,9 @@ struct BkTrack {
 };
 
 // fp32 fast path: bucket % 8 == 0, u32-indexable slice.  4 elems/lane, two
-// independent chains, no division, no per-access branches.
+// independent chains, no division, no per-access branches.  (U=4 in the
+// lean kernel: VGPR 56->106 for a wash — the fifth occupancy-vs-ILP
+// datapoint on this kernel family, all negative.)
 template <int BITS, bool AL16, bool HAVE>
 __device__ void deq_f32_fast(const DequantDesc& d,
                              const uint8_t* __restrict__ in0,
@@ -1097,7 +1099,7 @@ __device__ void deq_f32_fast(const DequantDesc& d,
     bk[u].init(static_cast<uint32_t>(t0 + u * stride),
                static_cast<uint32_t>(U * stride), B4);
   int64_t w = t0;
-  for (; w + stride < full_subs; w += U * stride) {
+  for (; w + (U - 1) * stride < full_subs; w += U * stride) {
     uint32_t v[U][4];
     float* outp[U];
 #pragma unroll
