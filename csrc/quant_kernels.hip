@@ -358,7 +358,11 @@ struct QuantRun {
   int slice_idx;
 
   __device__ __forceinline__ void load(Stash& s, int64_t lb) const {
-    const T* in = base + lb * (int64_t)(ngroups * 8);
+    load_from(s, base + lb * (int64_t)(ngroups * 8));
+  }
+
+  // arbitrary 16B-aligned source (global or LDS via generic pointer)
+  __device__ __forceinline__ void load_from(Stash& s, const T* in) const {
 #pragma unroll
     for (int k = 0; k < G; k++) {
       const int g = lane + k * kWave;
