@@ -141,3 +141,42 @@ def test_skip_incomplete_sim(ws):
         assert torch.equal(o, out[0])
     err = (out[0] - exact).abs()
     assert err.max().item() < 2 * bucket / 15 * ws * (ws + 1)
+
+
+@pytest.mark.parametrize("nodes,local", [(2, 2), (2, 4), (4, 2)])
+def test_hierarchical_compose_cpu(nodes, local):
+    """The hierarchical composition (per-node SRA -> cross reduction over
+    node results -> broadcast quantize round trip) stays within the stacked
+    analytic bounds and is rank-identical — the CPU oracle the GPU
+    hierarchical loopback test compares against, checked standalone."""
+    from torch_cgx_amd.ops import golden
+    torch.manual_seed(nodes * 10 + local)
+    bits, bucket, n = 4, 512, 8192
+    ws = nodes * local
+    tensors = [torch.randn(n) for _ in range(ws)]
+    exact = sum(tensors)
+
+    node_res = []
+    for nd in range(nodes):
+        r = sra_sim.sra_allreduce(
+            [tensors[nd * local + k].clone() for k in range(local)],
+            [n], [(bits, bucket)])
+        node_res.append(r[0])
+    if nodes > 2:
+        cross = sra_sim.ring_allreduce([t.clone() for t in node_res],
+                                       [n], [(bits, bucket)])
+    else:
+        cross = sra_sim.sra_allreduce([t.clone() for t in node_res],
+                                      [n], [(bits, bucket)])
+    final = golden.dequantize(
+        golden.quantize(cross[0], bits, bucket, rand=0.5), n,
+        torch.float32, bits, bucket)
+
+    # stacked bound: intra (local ranks) + cross (nodes, larger magnitudes)
+    # + one broadcast quantization of the full sum
+    unit_scale = 2 * bucket / ((1 << bits) - 1)
+    bound = unit_scale * (local * (local + 1)
+                          + local * nodes * (nodes + 1)
+                          + ws + 1)
+    err = (final - exact).abs().max().item()
+    assert err < bound, (nodes, local, err, bound)
